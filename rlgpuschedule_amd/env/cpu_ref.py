@@ -1,0 +1,273 @@
+"""Pure-Python reference implementation of the cluster-scheduling env.
+
+This is the ORACLE: the batched HIP env (csrc/env_step.hip) is differential-
+tested against this implementation for bit-exact integer state and
+atol=1e-5 observations. Semantics: docs/ENV_SPEC.md. Keep this file simple
+and obviously-correct rather than fast; the fast paths are the HIP kernel
+and (for CPU-only configs) the vectorized numpy env.
+"""
+from __future__ import annotations
+
+import math
+
+import numpy as np
+
+from ..traces.synthetic import JobTrace
+from .spec import EnvSpec
+
+NO_EVENT = np.int64(0x7FFFFFFFFFFFFFFF)
+
+
+def _lognorm(x_us: int) -> np.float32:
+    # float32 math end-to-end, mirrored by the HIP kernel (log1pf).
+    return np.float32(math.log1p(np.float32(x_us) / np.float32(1e6)) / np.float32(10.0))
+
+
+class ClusterEnvRef:
+    """Single-environment reference implementation (integer-tick DES)."""
+
+    def __init__(self, spec: EnvSpec, trace: JobTrace, job_lo: int = 0,
+                 job_hi: int | None = None):
+        self.spec = spec
+        self.trace = trace
+        self.job_lo = int(job_lo)
+        self.job_hi = int(len(trace) if job_hi is None else job_hi)
+        assert self.job_lo < self.job_hi <= len(trace)
+        trace.validate(spec.max_job_gpus, self.job_lo, self.job_hi)
+        self.base = int(trace.arrival_us[self.job_lo])
+        self.reset()
+
+    # -- helpers -----------------------------------------------------------
+    def _arrival(self, j: int) -> int:
+        return int(self.trace.arrival_us[j]) - self.base
+
+    def _work(self, j: int, pend_pos: int | None = None) -> int:
+        if pend_pos is not None and self.pend_override[pend_pos] > 0:
+            return int(self.pend_override[pend_pos])
+        return int(self.trace.work_us[j])
+
+    def first_fit(self, g: int) -> list[tuple[int, int]] | None:
+        """Deterministic first-fit: returns alloc pairs or None."""
+        pairs: list[tuple[int, int]] = []
+        left = g
+        for n in range(self.spec.num_nodes):
+            free = int(self.node_free[n])
+            if free > 0:
+                take = min(free, left)
+                if len(pairs) == self.spec.max_span:
+                    return None
+                pairs.append((n, take))
+                left -= take
+                if left == 0:
+                    return pairs
+        return None
+
+    def _next_event(self) -> int:
+        t = NO_EVENT
+        for r in range(self.spec.run_cap):
+            if self.run_valid[r] and self.run_finish[r] < t:
+                t = self.run_finish[r]
+        if self.cursor < self.job_hi:
+            t = min(t, self._arrival(self.cursor))
+        return int(t)
+
+    def _admit(self) -> None:
+        while (self.cursor < self.job_hi
+               and self._arrival(self.cursor) <= self.clock
+               and self.pend_count < self.spec.queue_cap):
+            self.pend_job[self.pend_count] = self.cursor
+            self.pend_override[self.pend_count] = 0
+            self.pend_count += 1
+            self.cursor += 1
+
+    # -- API ----------------------------------------------------------------
+    def reset(self) -> np.ndarray:
+        s = self.spec
+        self.clock = 0
+        self.cursor = self.job_lo
+        self.node_free = np.full(s.num_nodes, s.gpus_per_node, np.int32)
+        self.pend_job = np.zeros(s.queue_cap, np.int64)
+        self.pend_override = np.zeros(s.queue_cap, np.int64)  # preempted remaining work
+        self.pend_count = 0
+        self.run_valid = np.zeros(s.run_cap, bool)
+        self.run_job = np.zeros(s.run_cap, np.int64)
+        self.run_gpus = np.zeros(s.run_cap, np.int32)
+        self.run_finish = np.zeros(s.run_cap, np.int64)
+        self.run_start = np.zeros(s.run_cap, np.int64)
+        self.run_alloc: list[list[tuple[int, int]]] = [[] for _ in range(s.run_cap)]
+        self.run_count = 0
+        self.decisions = 0
+        self.completed = 0
+        self.jct_sum_us = 0
+        self.reward_sum = np.float32(0.0)
+        self._admit()
+        return self.observe()
+
+    def _free_total(self) -> int:
+        return int(self.node_free.sum())
+
+    def _place(self, pend_pos: int) -> bool:
+        j = int(self.pend_job[pend_pos])
+        g = int(self.trace.gpus[j])
+        pairs = self.first_fit(g)
+        if pairs is None:
+            return False
+        for n, c in pairs:
+            self.node_free[n] -= c
+        slot = next(r for r in range(self.spec.run_cap) if not self.run_valid[r])
+        self.run_valid[slot] = True
+        self.run_job[slot] = j
+        self.run_gpus[slot] = g
+        self.run_start[slot] = self.clock
+        self.run_finish[slot] = self.clock + self._work(j, pend_pos)
+        self.run_alloc[slot] = pairs
+        self.run_count += 1
+        # remove pending position
+        for p in range(pend_pos, self.pend_count - 1):
+            self.pend_job[p] = self.pend_job[p + 1]
+            self.pend_override[p] = self.pend_override[p + 1]
+        self.pend_count -= 1
+        return True
+
+    def _release(self, slot: int) -> None:
+        for n, c in self.run_alloc[slot]:
+            self.node_free[n] += c
+        self.run_valid[slot] = False
+        self.run_alloc[slot] = []
+        self.run_count -= 1
+
+    def _advance(self) -> np.float32:
+        t_next = self._next_event()
+        if t_next == NO_EVENT:
+            t_next = self.clock + self.spec.idle_tick_us
+        dt = t_next - self.clock
+        n_sys = self.pend_count + self.run_count
+        reward = np.float32(-(np.float32(n_sys) * np.float32(dt))
+                            / np.float32(self.spec.time_scale_us))
+        self.clock = t_next
+        for r in range(self.spec.run_cap):
+            if self.run_valid[r] and self.run_finish[r] <= self.clock:
+                j = int(self.run_job[r])
+                self.completed += 1
+                self.jct_sum_us += self.clock - self._arrival(j)
+                self._release(r)
+        self._admit()
+        return reward
+
+    def _migrate(self) -> bool:
+        """Gandiva MIGRATE: re-place the widest-span running job. False if
+        nothing is running (degrade to NOOP)."""
+        if self.run_count == 0:
+            return False
+        spans = [(len(self.run_alloc[r]), r) for r in range(self.spec.run_cap)
+                 if self.run_valid[r]]
+        span, slot = max(spans, key=lambda t: (t[0], -t[1]))
+        # free, then first-fit re-place (may land identically)
+        pairs_old = self.run_alloc[slot]
+        for n, c in pairs_old:
+            self.node_free[n] += c
+        pairs_new = self.first_fit(int(self.run_gpus[slot]))
+        if pairs_new is None:
+            # span-constrained first-fit can fail even after freeing (the
+            # freed cluster may be more fragmented than at placement time):
+            # restore and degrade to NOOP.
+            for n, c in pairs_old:
+                self.node_free[n] -= c
+            return False
+        for n, c in pairs_new:
+            self.node_free[n] -= c
+        self.run_alloc[slot] = pairs_new
+        return True
+
+    def _timeslice(self) -> bool:
+        """Gandiva TIMESLICE: preempt the longest-elapsed running job to the
+        pending tail. False if nothing running or pending full."""
+        if self.run_count == 0 or self.pend_count >= self.spec.queue_cap:
+            return False
+        el = [(self.clock - int(self.run_start[r]), -r) for r in range(self.spec.run_cap)
+              if self.run_valid[r]]
+        _, negr = max(el)
+        slot = -negr
+        remaining = max(1, int(self.run_finish[slot]) - self.clock)
+        self.pend_job[self.pend_count] = self.run_job[slot]
+        self.pend_override[self.pend_count] = remaining
+        self.pend_count += 1
+        self._release(slot)
+        return True
+
+    def step(self, action: int) -> tuple[np.ndarray, np.float32, bool, dict]:
+        s = self.spec
+        reward = np.float32(0.0)
+        a = int(action)
+        did = False
+        if a < s.queue_visible:
+            if a < self.pend_count:
+                did = self._place(a)
+        elif s.gandiva and a == s.migrate_action:
+            did = self._migrate()
+        elif s.gandiva and a == s.timeslice_action:
+            did = self._timeslice()
+        if not did:
+            reward = self._advance()
+        self.decisions += 1
+        self.reward_sum = np.float32(self.reward_sum + reward)
+        done = ((self.cursor == self.job_hi and self.pend_count == 0
+                 and self.run_count == 0)
+                or self.decisions >= s.horizon)
+        info = {"action_mask": self.action_mask()}
+        if done:
+            info["episode"] = {
+                "return": float(self.reward_sum),
+                "decisions": self.decisions,
+                "completed": self.completed,
+                "mean_jct_us": (self.jct_sum_us / self.completed
+                                if self.completed else 0.0),
+            }
+        return self.observe(), reward, done, info
+
+    def action_mask(self) -> np.ndarray:
+        s = self.spec
+        m = np.zeros(s.num_actions, bool)
+        for k in range(min(s.queue_visible, self.pend_count)):
+            j = int(self.pend_job[k])
+            m[k] = self.first_fit(int(self.trace.gpus[j])) is not None
+        m[s.noop_action] = True
+        if s.gandiva:
+            m[s.migrate_action] = self.run_count > 0
+            m[s.timeslice_action] = (self.run_count > 0
+                                     and self.pend_count < s.queue_cap)
+        return m
+
+    def observe(self) -> np.ndarray:
+        s = self.spec
+        obs = np.zeros(s.obs_dim, np.float32)
+        C = s.total_gpus
+        obs[0] = np.float32(self._free_total()) / np.float32(C)
+        obs[1] = np.float32(self.pend_count) / np.float32(s.queue_cap)
+        obs[2] = np.float32(self.run_count) / np.float32(s.run_cap)
+        t_next = self._next_event()
+        obs[3] = _lognorm(t_next - self.clock) if t_next != NO_EVENT else np.float32(0)
+        for k in range(min(s.queue_visible, self.pend_count)):
+            j = int(self.pend_job[k])
+            g = int(self.trace.gpus[j])
+            b = 4 + 5 * k
+            obs[b + 0] = 1.0
+            obs[b + 1] = np.float32(g) / np.float32(C)
+            obs[b + 2] = _lognorm(self.clock - self._arrival(j))
+            obs[b + 3] = _lognorm(self._work(j, k))
+            obs[b + 4] = np.float32(1.0 if self.first_fit(g) is not None else 0.0)
+        return obs
+
+    # integer-state snapshot for differential tests against the HIP env
+    def state_fingerprint(self) -> dict:
+        return {
+            "clock": self.clock,
+            "cursor": self.cursor,
+            "pend_count": self.pend_count,
+            "run_count": self.run_count,
+            "node_free": self.node_free.copy(),
+            "pend_job": self.pend_job[: self.pend_count].copy(),
+            "completed": self.completed,
+            "jct_sum_us": self.jct_sum_us,
+            "decisions": self.decisions,
+        }
