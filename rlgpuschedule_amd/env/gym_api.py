@@ -69,11 +69,16 @@ class CpuVectorEnv:
     """Batched CPU env over per-env trace windows, with auto-reset."""
 
     def __init__(self, spec: EnvSpec, num_envs: int, trace_kind: str = "philly",
-                 jobs_per_env: int = 128, seed: int = 0, **trace_kw):
+                 jobs_per_env: int = 128, seed: int = 0,
+                 pool: JobTrace | None = None, job_lo=None, job_hi=None,
+                 **trace_kw):
         self.spec = spec
         self.num_envs = num_envs
-        pool, lo, hi = make_pool(trace_kind, num_envs, jobs_per_env, seed,
-                                 spec.max_job_gpus, **trace_kw)
+        if pool is None:
+            pool, lo, hi = make_pool(trace_kind, num_envs, jobs_per_env, seed,
+                                     spec.max_job_gpus, **trace_kw)
+        else:
+            lo, hi = job_lo, job_hi
         self.pool = pool
         self.job_lo, self.job_hi = lo, hi
         self.envs = [ClusterEnvRef(spec, pool, int(lo[e]), int(hi[e]))
