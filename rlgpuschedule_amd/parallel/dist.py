@@ -1,0 +1,94 @@
+"""Data parallelism over torch.distributed — RCCL over xGMI on MI355X.
+
+One process per GPU (torch.distributed.run). Backend "nccl" IS RCCL on
+ROCm; CPU tests use gloo. Gradient sync is ONE fused flat all-reduce per
+update (xGMI is point-to-point, 7 links/GPU: for the small MLP/GRU payloads
+here, <10 MB, latency dominates — a single flat buffer beats bucketing).
+"""
+from __future__ import annotations
+
+import os
+from contextlib import contextmanager
+
+import torch
+import torch.distributed as dist
+
+
+def init_from_env() -> tuple[int, int, int]:
+    """Returns (rank, world_size, local_rank); initializes the process group
+    when launched under torch.distributed.run, else single-process."""
+    if "RANK" not in os.environ or dist.is_initialized():
+        if dist.is_initialized():
+            return dist.get_rank(), dist.get_world_size(), _local_rank()
+        return 0, 1, 0
+    backend = "nccl" if torch.cuda.is_available() else "gloo"
+    dist.init_process_group(backend=backend)
+    lr = _local_rank()
+    if torch.cuda.is_available():
+        torch.cuda.set_device(lr)
+    return dist.get_rank(), dist.get_world_size(), lr
+
+
+def _local_rank() -> int:
+    return int(os.environ.get("LOCAL_RANK", 0))
+
+
+def world_size() -> int:
+    return dist.get_world_size() if dist.is_initialized() else 1
+
+
+def barrier():
+    if dist.is_initialized():
+        dist.barrier()
+
+
+def broadcast_params(model: torch.nn.Module, src: int = 0):
+    if not dist.is_initialized():
+        return
+    for p in model.parameters():
+        dist.broadcast(p.data, src)
+
+
+class FlatGradAllreduce:
+    """Fused flat-buffer gradient all-reduce (mean) — one RCCL call/update.
+
+    Grads are copied into one contiguous flat buffer, all-reduced, copied
+    back. For MLP/GRU policies (<10 MB of grads) this is latency-optimal on
+    xGMI; overlap with backward is unnecessary at this size.
+    """
+
+    def __init__(self, model: torch.nn.Module):
+        self.params = [p for p in model.parameters() if p.requires_grad]
+        numel = sum(p.numel() for p in self.params)
+        dev = self.params[0].device
+        self.flat = torch.zeros(numel, dtype=torch.float32, device=dev)
+
+    def __call__(self, model=None):
+        if not dist.is_initialized() or dist.get_world_size() == 1:
+            return
+        off = 0
+        for p in self.params:
+            n = p.numel()
+            if p.grad is not None:
+                self.flat[off:off + n].copy_(p.grad.detach().reshape(-1))
+            else:
+                self.flat[off:off + n].zero_()
+            off += n
+        self.flat.div_(dist.get_world_size())
+        dist.all_reduce(self.flat)
+        off = 0
+        for p in self.params:
+            n = p.numel()
+            if p.grad is None:
+                p.grad = torch.zeros_like(p)
+            p.grad.copy_(self.flat[off:off + n].view_as(p))
+            off += n
+
+
+def allreduce_scalar(x: float, op: str = "max") -> float:
+    if not dist.is_initialized():
+        return x
+    t = torch.tensor([x], dtype=torch.float64,
+                     device="cuda" if dist.get_backend() == "nccl" else "cpu")
+    dist.all_reduce(t, op=dist.ReduceOp.MAX if op == "max" else dist.ReduceOp.SUM)
+    return float(t.item())

@@ -1,0 +1,132 @@
+"""End-to-end training driver wiring env + policy + learner + DP.
+
+The five named configs mirror BASELINE.json's config list.
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass, field, replace
+
+import torch
+
+from .agents.ppo import A2CConfig, A2CLearner, PPOConfig, PPOLearner
+from .agents.rollout import Collector, TorchEnvAdapter
+from .env.gym_api import CpuVectorEnv
+from .env.spec import EnvSpec, cluster16, cluster64, cluster256, cluster1024
+from .models.policy import GRUPolicy, MLPPolicy
+from .parallel.dist import FlatGradAllreduce, broadcast_params, init_from_env
+
+
+@dataclass
+class TrainConfig:
+    name: str = "c2_ppo_mlp"
+    spec: EnvSpec = field(default_factory=cluster64)
+    num_envs: int = 1024          # per GPU
+    rollout_T: int = 64
+    algo: str = "ppo"             # ppo | a2c | impala
+    policy: str = "mlp"           # mlp | gru
+    hidden: int = 256
+    trace_kind: str = "philly"
+    jobs_per_env: int = 256
+    seed: int = 0
+    iterations: int = 100
+    device: str = "auto"
+    ppo: PPOConfig = field(default_factory=PPOConfig)
+    a2c: A2CConfig = field(default_factory=A2CConfig)
+    use_hip_env: bool = True      # False -> CPU reference env
+
+
+CONFIGS = {
+    # 16-GPU simulated cluster, random-policy agent on CPU Gym env
+    "c1_cpu_random": TrainConfig(
+        name="c1_cpu_random", spec=cluster16(), num_envs=8, rollout_T=32,
+        algo="random", policy="mlp", use_hip_env=False, jobs_per_env=64),
+    # 64-GPU Philly-trace sim, PPO MLP, 256 envs, 1 GPU
+    "c2_ppo_mlp": TrainConfig(
+        name="c2_ppo_mlp", spec=cluster64(), num_envs=256, rollout_T=64,
+        algo="ppo", policy="mlp"),
+    # 256-GPU sim, PPO GRU, 2048 envs, DP=8
+    "c3_ppo_gru": TrainConfig(
+        name="c3_ppo_gru", spec=cluster256(), num_envs=2048, rollout_T=64,
+        algo="ppo", policy="gru"),
+    # Gandiva actions + elastic jobs, A2C
+    "c4_gandiva_a2c": TrainConfig(
+        name="c4_gandiva_a2c", spec=cluster64(gandiva=True), num_envs=1024,
+        rollout_T=64, algo="a2c", policy="mlp"),
+    # 1024-GPU Poisson trace, IMPALA-style async actor/learner
+    "c5_impala": TrainConfig(
+        name="c5_impala", spec=cluster1024(), num_envs=2048, rollout_T=64,
+        algo="impala", policy="mlp", trace_kind="poisson"),
+}
+
+
+def resolve_device(cfg: TrainConfig) -> torch.device:
+    if cfg.device != "auto":
+        return torch.device(cfg.device)
+    return torch.device("cuda" if torch.cuda.is_available() else "cpu")
+
+
+def make_env(cfg: TrainConfig, device: torch.device, rank: int = 0):
+    seed = cfg.seed * 7919 + rank
+    if cfg.use_hip_env and device.type == "cuda":
+        from .env.hip_env import HipVectorEnv
+        return HipVectorEnv(cfg.spec, cfg.num_envs, trace_kind=cfg.trace_kind,
+                            jobs_per_env=cfg.jobs_per_env, seed=seed)
+    venv = CpuVectorEnv(cfg.spec, cfg.num_envs, trace_kind=cfg.trace_kind,
+                        jobs_per_env=cfg.jobs_per_env, seed=seed)
+    return TorchEnvAdapter(venv, device)
+
+
+def make_policy(cfg: TrainConfig, device: torch.device):
+    cls = GRUPolicy if cfg.policy == "gru" else MLPPolicy
+    pol = cls(cfg.spec.obs_dim, cfg.spec.num_actions, cfg.hidden)
+    return pol.to(device)
+
+
+def make_learner(cfg: TrainConfig, policy, grad_sync=None):
+    if cfg.algo == "ppo":
+        return PPOLearner(policy, cfg.ppo, grad_sync)
+    if cfg.algo == "a2c":
+        return A2CLearner(policy, cfg.a2c, grad_sync)
+    raise ValueError(f"unknown algo {cfg.algo}")
+
+
+def train(cfg: TrainConfig, iterations: int | None = None,
+          log_fn=print) -> dict:
+    """Synchronous trainer (ppo/a2c). IMPALA lives in agents/impala.py."""
+    rank, world, local = init_from_env()
+    device = resolve_device(cfg)
+    if device.type == "cuda":
+        device = torch.device(f"cuda:{local}")
+    torch.manual_seed(cfg.seed * 31 + rank)
+    if cfg.algo == "impala":
+        from .agents.impala import train_impala
+        return train_impala(cfg, iterations=iterations, log_fn=log_fn)
+    env = make_env(cfg, device, rank)
+    policy = make_policy(cfg, device)
+    broadcast_params(policy)
+    grad_sync = FlatGradAllreduce(policy) if world > 1 else None
+    collector = Collector(env, policy, cfg.rollout_T, device,
+                          seed=cfg.seed * 101 + rank)
+    if cfg.algo == "random":
+        learner = None
+    else:
+        learner = make_learner(cfg, policy, grad_sync)
+    iters = iterations if iterations is not None else cfg.iterations
+    stats = {}
+    for it in range(iters):
+        buf = collector.collect()
+        if learner is not None:
+            stats = learner.update(buf)
+        if rank == 0 and log_fn and (it % 10 == 0 or it == iters - 1):
+            log_fn(f"[{cfg.name}] iter {it}: " +
+                   " ".join(f"{k}={v:.4f}" for k, v in stats.items()))
+    out = dict(stats)
+    if hasattr(env, "episode_stats"):
+        out["episode_stats"] = env.episode_stats()
+    elif hasattr(env, "venv"):
+        st = env.venv.stats
+        out["episode_stats"] = {
+            "episodes": st.count, "mean_return": st.mean_return(),
+            "mean_jct_us": st.mean_jct_us(), "completed": st.completed_sum,
+        }
+    return out
