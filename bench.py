@@ -70,7 +70,11 @@ def main():
     # Throughput-oriented PPO: 1 epoch x 8 minibatches (documented in config)
     cfg = PPOConfig(epochs=1, minibatches=8)
     learner = PPOLearner(policy, cfg, grad_sync)
-    collector = Collector(env, policy, T, device, seed=555 + rank)
+    if has_gpu:
+        from rlgpuschedule_amd.agents.rollout_gpu import GpuRolloutEngine
+        collector = GpuRolloutEngine(env, policy, T, seed=555 + rank)
+    else:
+        collector = Collector(env, policy, T, device, seed=555 + rank)
 
     def one_iter():
         buf = collector.collect()

@@ -16,6 +16,16 @@ from ..ops.gae_ref import gae_advantages
 from .rollout import RolloutBuffer
 
 
+def compute_gae(buf: RolloutBuffer, gamma: float, lam: float):
+    """HIP segmented-scan kernel on GPU, torch reference on CPU."""
+    if buf.reward.is_cuda:
+        from ..ops.nn import gae_scan
+        return gae_scan(buf.reward, buf.value, buf.done, buf.last_value,
+                        gamma, lam)
+    return gae_advantages(buf.reward, buf.value, buf.done, buf.last_value,
+                          gamma, lam)
+
+
 @dataclass
 class PPOConfig:
     lr: float = 3e-4
@@ -39,8 +49,7 @@ class PPOLearner:
 
     def update(self, buf: RolloutBuffer) -> dict:
         cfg = self.cfg
-        adv, ret = gae_advantages(buf.reward, buf.value, buf.done,
-                                  buf.last_value, cfg.gamma, cfg.lam)
+        adv, ret = compute_gae(buf, cfg.gamma, cfg.lam)
         stats = {"pi_loss": 0.0, "v_loss": 0.0, "entropy": 0.0, "clipfrac": 0.0}
         nup = 0
         if self.policy.is_recurrent:
@@ -161,8 +170,7 @@ class A2CLearner:
 
     def update(self, buf: RolloutBuffer) -> dict:
         cfg = self.cfg
-        adv, ret = gae_advantages(buf.reward, buf.value, buf.done,
-                                  buf.last_value, cfg.gamma, cfg.lam)
+        adv, ret = compute_gae(buf, cfg.gamma, cfg.lam)
         T, E = buf.reward.shape
         if self.policy.is_recurrent:
             hx = buf.h0

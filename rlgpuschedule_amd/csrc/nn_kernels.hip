@@ -1,0 +1,249 @@
+// Hand-written CDNA4 kernels for the RL hot path: GAE scan + fused MLP actor.
+//
+// actor_mlp_kernel — ONE launch per rollout step computes, for all E envs:
+//     h1 = tanh(obs @ W1t + b1)      [E,H]   (MFMA f32 16x16x4)
+//     h2 = tanh(h1 @ W2t + b2)       [E,H]
+//     head = h2 @ Wht + bh           [E,32]  (cols [0,A) logits, col A value)
+//     masked softmax -> inverse-CDF sample with caller-provided uniforms
+//   Layout: 16 env rows per 256-thread workgroup (4 waves); each wave owns
+//   64 output columns per chunk; activations live in LDS (padded stride
+//   H+1 to break bank conflicts); weights stream from global and stay
+//   L2-resident (W2 fp32 = 256 KB/XCD-L2). MFMA f32_16x16x4 is exact f32
+//   at the 157 TF vector rate (cdna_hip_programming.md §3) — numerics match
+//   the torch fp32 reference to reassociation tolerance.
+//   Weight tensors are the TRANSPOSED copies maintained by
+//   models/hip_actor.py from the torch parameters ([in,out] layout so the
+//   B-fragment read B[k][n] is coalesced: 16 consecutive lanes read 16
+//   consecutive columns).
+//
+// gae_kernel — segmented reverse scan over [T,E]: one thread per env,
+//   time-step loads coalesced across the wave (layout [t][e]).
+//
+// Numerics tests: tests/test_nn_kernels.py vs torch fp32 reference.
+
+#include <hip/hip_runtime.h>
+#include <stdint.h>
+
+#define WAVE 64
+
+typedef float f32x4 __attribute__((ext_vector_type(4)));
+
+namespace {
+
+// ---------------------------------------------------------------- GAE scan
+struct GaeArgs {
+  int64_t T, E;
+  float gamma, lam;
+  const float* reward;      // [T*E]
+  const float* value;       // [T*E]
+  const uint8_t* done;      // [T*E]
+  const float* last_value;  // [E]
+  float* adv;               // [T*E]
+  float* ret;               // [T*E]
+};
+
+__global__ __launch_bounds__(256) void gae_kernel(GaeArgs a) {
+  int64_t e = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (e >= a.E) return;
+  float lastg = 0.0f;
+  float next_v = a.last_value[e];
+  for (int64_t t = a.T - 1; t >= 0; --t) {
+    int64_t ix = t * a.E + e;
+    float nd = 1.0f - (float)a.done[ix];
+    float v = a.value[ix];
+    float delta = a.reward[ix] + a.gamma * next_v * nd - v;
+    lastg = delta + a.gamma * a.lam * nd * lastg;
+    a.adv[ix] = lastg;
+    a.ret[ix] = lastg + v;
+    next_v = v;
+  }
+}
+
+// ------------------------------------------------------- fused MLP actor
+struct ActorArgs {
+  int64_t E, D, H, A;       // envs, obs_dim, hidden, num_actions
+  int64_t value_only;       // 1: only write value (bootstrap pass)
+  const float* obs;         // [E,D]
+  const uint8_t* mask;      // [E,A]
+  const float* u;           // [E] uniforms (ignored if value_only)
+  const float* wt1;         // [D,H] transposed W1
+  const float* b1;          // [H]
+  const float* wt2;         // [H,H]
+  const float* b2;          // [H]
+  const float* wth;         // [H,32] logits cols [0,A), value col A, rest 0
+  const float* bh;          // [32]
+  int64_t* actions;         // [E]
+  float* logp;              // [E]
+  float* value;             // [E]
+};
+
+#define ROWS 16      // env rows per workgroup
+#define HEADC 32     // head columns (A+1 <= 32)
+
+// One 64-column GEMM chunk for one wave: C[16 x 64] += A[16 x K] * B[K x 64].
+// A is an LDS tile with row stride lda; B global row-major [K][ldb].
+// Result written to LDS out (row stride ldo) through act_tanh epilogue.
+__device__ __forceinline__ void gemm_chunk_16x64(
+    const float* __restrict__ a_lds, int lda, int K,
+    const float* __restrict__ b_glb, int ldb, int col0,
+    const float* __restrict__ bias, bool do_tanh,
+    float* out_lds, int ldo, int lane) {
+  f32x4 acc[4] = {};
+  const int ar = lane & 15;        // A row
+  const int kk = lane >> 4;        // A/B k sub-index (0..3)
+  const int bc = lane & 15;        // B col within 16-group
+  for (int k0 = 0; k0 < K; k0 += 4) {
+    float av = a_lds[ar * lda + k0 + kk];
+    const float* brow = b_glb + (int64_t)(k0 + kk) * ldb + col0;
+    #pragma unroll
+    for (int n = 0; n < 4; ++n) {
+      float bv = brow[n * 16 + bc];
+      acc[n] = __builtin_amdgcn_mfma_f32_16x16x4f32(av, bv, acc[n], 0, 0, 0);
+    }
+  }
+  // C/D map (16x16): col = lane&15, row = (lane>>4)*4 + i
+  const int crow = (lane >> 4) * 4;
+  const int ccol = lane & 15;
+  #pragma unroll
+  for (int n = 0; n < 4; ++n) {
+    #pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      int col = col0 + n * 16 + ccol;
+      float vv = acc[n][i] + bias[col];
+      if (do_tanh) vv = tanhf(vv);
+      out_lds[(crow + i) * ldo + col] = vv;
+    }
+  }
+}
+
+__global__ __launch_bounds__(256) void actor_mlp_kernel(ActorArgs p) {
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int w = tid >> 6;
+  const int64_t row0 = (int64_t)blockIdx.x * ROWS;
+  const int D = (int)p.D, H = (int)p.H, A = (int)p.A;
+  const int lda_o = D + 1, lda_h = H + 1;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* sobs = (float*)smem;                       // [ROWS][D+1]
+  float* h1 = sobs + ROWS * lda_o;                  // [ROWS][H+1]
+  float* h2 = h1 + ROWS * lda_h;                    // [ROWS][H+1]
+  float* shead = h2 + ROWS * lda_h;                 // [ROWS][HEADC+1]
+
+  // stage obs tile (zero-pad rows past E and the k-pad column)
+  for (int ix = tid; ix < ROWS * lda_o; ix += 256) {
+    int r = ix / lda_o, d = ix % lda_o;
+    float vv = 0.0f;
+    if (d < D && row0 + r < p.E) vv = p.obs[(row0 + r) * D + d];
+    sobs[r * lda_o + d] = vv;
+  }
+  __syncthreads();
+
+  // layer 1: obs[16,D] @ wt1[D,H]
+  for (int c0 = w * 64; c0 < H; c0 += 256)
+    gemm_chunk_16x64(sobs, lda_o, D, p.wt1, H, c0, p.b1, true, h1, lda_h, lane);
+  // zero h1 pad column (k-loop over H reads index H? no: K=H, k0+kk < H+4?
+  // K=H is a multiple of 4 so k0+kk <= H-1; pad col never read as A)
+  __syncthreads();
+
+  // layer 2: h1[16,H] @ wt2[H,H]
+  for (int c0 = w * 64; c0 < H; c0 += 256)
+    gemm_chunk_16x64(h1, lda_h, H, p.wt2, H, c0, p.b2, true, h2, lda_h, lane);
+  __syncthreads();
+
+  // head: h2[16,H] @ wth[H,32] — wave 0 only (tiny)
+  if (w == 0) {
+    f32x4 acc[2] = {};
+    const int ar = lane & 15, kk = lane >> 4, bc = lane & 15;
+    for (int k0 = 0; k0 < H; k0 += 4) {
+      float av = h2[ar * lda_h + k0 + kk];
+      const float* brow = p.wth + (int64_t)(k0 + kk) * HEADC;
+      #pragma unroll
+      for (int n = 0; n < 2; ++n) {
+        float bv = brow[n * 16 + bc];
+        acc[n] = __builtin_amdgcn_mfma_f32_16x16x4f32(av, bv, acc[n], 0, 0, 0);
+      }
+    }
+    const int crow = (lane >> 4) * 4, ccol = lane & 15;
+    #pragma unroll
+    for (int n = 0; n < 2; ++n)
+      #pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        int col = n * 16 + ccol;
+        shead[(crow + i) * (HEADC + 1) + col] = acc[n][i] + p.bh[col];
+      }
+  }
+  __syncthreads();
+
+  // per-row masked softmax + inverse-CDF sample: lane r of wave 0 owns row r
+  if (w == 0 && lane < ROWS) {
+    int64_t row = row0 + lane;
+    if (row < p.E) {
+      const float* hrow = shead + lane * (HEADC + 1);
+      p.value[row] = hrow[A];
+      if (!p.value_only) {
+        const uint8_t* mrow = p.mask + row * A;
+        float mx = -3.0e38f;
+        for (int i = 0; i < A; ++i)
+          if (mrow[i]) mx = fmaxf(mx, hrow[i]);
+        float sum = 0.0f;
+        for (int i = 0; i < A; ++i)
+          if (mrow[i]) sum += __expf(hrow[i] - mx);
+        float thresh = p.u[row] * sum;
+        float s = 0.0f;
+        int act = -1;
+        for (int i = 0; i < A; ++i) {
+          if (!mrow[i]) continue;
+          s += __expf(hrow[i] - mx);
+          act = i;                 // last valid as fallback
+          if (s >= thresh) break;
+        }
+        p.actions[row] = act;
+        p.logp[row] = (hrow[act] - mx) - __logf(sum);
+      }
+    }
+  }
+}
+
+}  // namespace
+
+extern "C" int rlsched_gae(const int64_t* ip, const double* fp,
+                           const uint64_t* bufs, void* stream) {
+  GaeArgs a;
+  a.T = ip[0]; a.E = ip[1];
+  a.gamma = (float)fp[0]; a.lam = (float)fp[1];
+  a.reward = (const float*)bufs[0];
+  a.value = (const float*)bufs[1];
+  a.done = (const uint8_t*)bufs[2];
+  a.last_value = (const float*)bufs[3];
+  a.adv = (float*)bufs[4];
+  a.ret = (float*)bufs[5];
+  dim3 grid((unsigned)((a.E + 255) / 256));
+  hipLaunchKernelGGL(gae_kernel, grid, dim3(256), 0, (hipStream_t)stream, a);
+  return (int)hipGetLastError();
+}
+
+extern "C" int rlsched_actor_mlp(const int64_t* ip, const uint64_t* bufs,
+                                 void* stream) {
+  ActorArgs p;
+  p.E = ip[0]; p.D = ip[1]; p.H = ip[2]; p.A = ip[3]; p.value_only = ip[4];
+  if (p.D % 4 || p.H % 64 || p.A + 1 > HEADC) return -1;
+  p.obs = (const float*)bufs[0];
+  p.mask = (const uint8_t*)bufs[1];
+  p.u = (const float*)bufs[2];
+  p.wt1 = (const float*)bufs[3];
+  p.b1 = (const float*)bufs[4];
+  p.wt2 = (const float*)bufs[5];
+  p.b2 = (const float*)bufs[6];
+  p.wth = (const float*)bufs[7];
+  p.bh = (const float*)bufs[8];
+  p.actions = (int64_t*)bufs[9];
+  p.logp = (float*)bufs[10];
+  p.value = (float*)bufs[11];
+  size_t lds = sizeof(float) * (ROWS * (p.D + 1) + 2 * ROWS * (p.H + 1)
+                                + ROWS * (HEADC + 1));
+  dim3 grid((unsigned)((p.E + ROWS - 1) / ROWS));
+  hipLaunchKernelGGL(actor_mlp_kernel, grid, dim3(256), lds,
+                     (hipStream_t)stream, p);
+  return (int)hipGetLastError();
+}

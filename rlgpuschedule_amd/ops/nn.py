@@ -1,0 +1,70 @@
+"""ctypes bindings for the HIP NN kernels (librlsched_nn.so): GAE scan and
+the fused MLP actor. Launches are hipGraph-capturable."""
+from __future__ import annotations
+
+import ctypes
+
+import numpy as np
+import torch
+
+from .build import build_lib
+
+_lib = None
+
+
+def load() -> ctypes.CDLL:
+    global _lib
+    if _lib is None:
+        so = build_lib("librlsched_nn.so")
+        lib = ctypes.CDLL(str(so))
+        lib.rlsched_gae.restype = ctypes.c_int
+        lib.rlsched_gae.argtypes = [ctypes.c_void_p] * 4
+        lib.rlsched_actor_mlp.restype = ctypes.c_int
+        lib.rlsched_actor_mlp.argtypes = [ctypes.c_void_p] * 3
+        _lib = lib
+    return _lib
+
+
+def _stream(t: torch.Tensor) -> int:
+    return torch.cuda.current_stream(t.device).cuda_stream
+
+
+def _ptrs(*tensors: torch.Tensor) -> np.ndarray:
+    return np.array([t.data_ptr() for t in tensors], dtype=np.uint64)
+
+
+def gae_scan(reward: torch.Tensor, value: torch.Tensor, done: torch.Tensor,
+             last_value: torch.Tensor, gamma: float, lam: float,
+             adv: torch.Tensor | None = None, ret: torch.Tensor | None = None):
+    """HIP GAE(λ) segmented reverse scan. reward/value [T,E] f32,
+    done [T,E] u8, last_value [E]. Returns (adv, ret)."""
+    T, E = reward.shape
+    assert done.dtype == torch.uint8
+    adv = torch.empty_like(reward) if adv is None else adv
+    ret = torch.empty_like(reward) if ret is None else ret
+    ip = np.array([T, E], np.int64)
+    fp = np.array([gamma, lam], np.float64)
+    bufs = _ptrs(reward, value, done, last_value, adv, ret)
+    rc = load().rlsched_gae(
+        ip.ctypes.data_as(ctypes.c_void_p), fp.ctypes.data_as(ctypes.c_void_p),
+        bufs.ctypes.data_as(ctypes.c_void_p), ctypes.c_void_p(_stream(reward)))
+    if rc != 0:
+        raise RuntimeError(f"rlsched_gae failed: {rc}")
+    return adv, ret
+
+
+def actor_mlp(obs, mask, u, wt1, b1, wt2, b2, wth, bh, actions, logp, value,
+              value_only: bool = False):
+    """Fused MLP policy forward + masked inverse-CDF sampling (one launch)."""
+    E, D = obs.shape
+    H = wt1.shape[1]
+    A = mask.shape[1]
+    ip = np.array([E, D, H, A, 1 if value_only else 0], np.int64)
+    bufs = _ptrs(obs, mask, u, wt1, b1, wt2, b2, wth, bh, actions, logp, value)
+    rc = load().rlsched_actor_mlp(
+        ip.ctypes.data_as(ctypes.c_void_p),
+        bufs.ctypes.data_as(ctypes.c_void_p), ctypes.c_void_p(_stream(obs)))
+    if rc != 0:
+        raise RuntimeError(f"rlsched_actor_mlp failed: {rc} "
+                           f"(check D%4==0, H%64==0, A<=31)")
+    return actions, logp, value
