@@ -69,11 +69,13 @@ def main():
     grad_sync = pdist.FlatGradAllreduce(policy) if world > 1 else None
     # Throughput-oriented PPO: 1 epoch x 8 minibatches (documented in config)
     cfg = PPOConfig(epochs=1, minibatches=8)
-    learner = PPOLearner(policy, cfg, grad_sync)
     if has_gpu:
+        from rlgpuschedule_amd.agents.ppo_gpu import FusedPPOLearner
         from rlgpuschedule_amd.agents.rollout_gpu import GpuRolloutEngine
+        learner = FusedPPOLearner(policy, cfg, grad_sync)
         collector = GpuRolloutEngine(env, policy, T, seed=555 + rank)
     else:
+        learner = PPOLearner(policy, cfg, grad_sync)
         collector = Collector(env, policy, T, device, seed=555 + rank)
 
     def one_iter():

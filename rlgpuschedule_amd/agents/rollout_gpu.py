@@ -35,10 +35,11 @@ class GpuRolloutEngine:
         self.done_last = torch.empty(E, dtype=torch.uint8, device=self.device)
         self.use_graph = use_graph
         self._graph = None
-        # initial reset: env writes obs/mask into its default buffers
+        # initial reset: env writes obs/mask into its default buffers;
+        # obs_last/mask_last always hold the NEXT rollout's start state
         env.reset()
-        self.buf.obs[0].copy_(env.b["obs"])
-        self.buf.mask[0].copy_(env.b["mask"])
+        self.obs_last.copy_(env.b["obs"])
+        self.mask_last.copy_(env.b["mask"])
 
     def _rollout_body(self):
         buf, env, T = self.buf, self.env, self.T
@@ -58,6 +59,10 @@ class GpuRolloutEngine:
     def collect(self) -> RolloutBuffer:
         self.actor.refresh()
         self.u_all.uniform_(generator=self.gen)
+        # rollout starts from the state the previous one ended in; the t=0
+        # slices must stay untouched afterwards (the learner reads them)
+        self.buf.obs[0].copy_(self.obs_last)
+        self.buf.mask[0].copy_(self.mask_last)
         if self.use_graph:
             if self._graph is None:
                 # warm once on a side stream, then capture
@@ -66,7 +71,7 @@ class GpuRolloutEngine:
                 with torch.cuda.stream(s):
                     self._rollout_body()
                 torch.cuda.current_stream().wait_stream(s)
-                # re-chain so the captured t=0 reads the post-warmup state
+                # warmup advanced the env: re-chain the replay's start state
                 self.buf.obs[0].copy_(self.obs_last)
                 self.buf.mask[0].copy_(self.mask_last)
                 torch.cuda.synchronize()
@@ -76,7 +81,4 @@ class GpuRolloutEngine:
             self._graph.replay()
         else:
             self._rollout_body()
-        # next iteration starts where this one ended
-        self.buf.obs[0].copy_(self.obs_last)
-        self.buf.mask[0].copy_(self.mask_last)
         return self.buf

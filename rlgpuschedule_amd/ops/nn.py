@@ -68,3 +68,29 @@ def actor_mlp(obs, mask, u, wt1, b1, wt2, b2, wth, bh, actions, logp, value,
         raise RuntimeError(f"rlsched_actor_mlp failed: {rc} "
                            f"(check D%4==0, H%64==0, A<=31)")
     return actions, logp, value
+
+
+def _load_ppo():
+    lib = load()
+    if not hasattr(lib, "_ppo_ready"):
+        lib.rlsched_ppo_loss.restype = ctypes.c_int
+        lib.rlsched_ppo_loss.argtypes = [ctypes.c_void_p] * 4
+        lib._ppo_ready = True
+    return lib
+
+
+def ppo_loss_fused(logits, value, actions, old_logp, old_value, adv, ret,
+                   clip, vf_coef, ent_coef, dlogits, dvalue, stats):
+    """Fused PPO objective forward+backward. `stats` [4] f32 is ACCUMULATED
+    (zero it first): [pi_loss_sum, v_loss_sum, entropy_sum, clipfrac_sum].
+    dlogits/dvalue get the loss gradients (mean-scaled)."""
+    B, A = logits.shape
+    ip = np.array([B, A], np.int64)
+    fp = np.array([clip, vf_coef, ent_coef, 1.0 / B], np.float64)
+    bufs = _ptrs(logits, value, actions, old_logp, old_value, adv, ret,
+                 dlogits, dvalue, stats)
+    rc = _load_ppo().rlsched_ppo_loss(
+        ip.ctypes.data_as(ctypes.c_void_p), fp.ctypes.data_as(ctypes.c_void_p),
+        bufs.ctypes.data_as(ctypes.c_void_p), ctypes.c_void_p(_stream(logits)))
+    if rc != 0:
+        raise RuntimeError(f"rlsched_ppo_loss failed: {rc}")
