@@ -1,0 +1,162 @@
+"""IMPALA-style async actor/learner with V-trace (BASELINE config 5).
+
+MI355X-native design: no RPC framework. On each GPU, the ACTOR is the
+graph-replayed rollout engine running on its own HIP stream with a stale
+behavior-policy mirror (HipActorMLP, refreshed every `behavior_sync`
+updates), double-buffered; the LEARNER consumes the previously-completed
+buffer on the default stream, correcting off-policyness with the V-trace
+HIP scan (csrc/nn_kernels.hip). Events order buffer reuse. Across the 8
+GPUs of a node, learners are data-parallel over RCCL (FlatGradAllreduce).
+
+CPU fallback (tests): sequential actor/learner with an explicitly stale
+policy copy, torch V-trace reference.
+"""
+from __future__ import annotations
+
+import copy
+from dataclasses import dataclass
+
+import torch
+
+from ..models.policy import evaluate_actions
+from ..ops.gae_ref import vtrace_reference
+from .rollout import Collector, RolloutBuffer
+
+
+@dataclass
+class ImpalaConfig:
+    lr: float = 3e-4
+    gamma: float = 0.99
+    lam: float = 1.0
+    rho_bar: float = 1.0
+    c_bar: float = 1.0
+    vf_coef: float = 0.5
+    ent_coef: float = 0.01
+    max_grad_norm: float = 0.5
+    behavior_sync: int = 1   # refresh behavior policy every N updates
+
+
+class VtraceLearner:
+    """Computes the IMPALA loss on a rollout collected by a (possibly
+    stale) behavior policy whose log-probs are stored in buf.logp."""
+
+    def __init__(self, policy, cfg: ImpalaConfig, grad_sync=None):
+        self.policy = policy
+        self.cfg = cfg
+        self.grad_sync = grad_sync
+        self.opt = torch.optim.Adam(policy.parameters(), lr=cfg.lr, eps=1e-5)
+
+    def update(self, buf: RolloutBuffer) -> dict:
+        cfg = self.cfg
+        T, E = buf.reward.shape
+        B = T * E
+        logits, values, _ = self.policy(buf.obs.reshape(B, -1),
+                                        buf.mask.reshape(B, -1), None)
+        lp_cur, entropy = evaluate_actions(logits, buf.actions.reshape(B))
+        values = values.reshape(T, E)
+        lp_cur = lp_cur.reshape(T, E)
+        with torch.no_grad():
+            v_det = values.detach()
+            if buf.reward.is_cuda:
+                from ..ops.nn import vtrace_scan
+                vs, pg_adv = vtrace_scan(
+                    buf.reward, v_det, buf.done, buf.last_value,
+                    lp_cur.detach(), buf.logp, cfg.gamma, cfg.rho_bar,
+                    cfg.c_bar, cfg.lam)
+            else:
+                vs, pg_adv = vtrace_reference(
+                    buf.reward, v_det, buf.done, buf.last_value,
+                    lp_cur.detach(), buf.logp, cfg.gamma, cfg.rho_bar,
+                    cfg.c_bar, cfg.lam)
+        pi_loss = -(pg_adv * lp_cur).mean()
+        v_loss = 0.5 * ((values - vs) ** 2).mean()
+        ent = entropy.mean()
+        loss = pi_loss + cfg.vf_coef * v_loss - cfg.ent_coef * ent
+        self.opt.zero_grad(set_to_none=True)
+        loss.backward()
+        if self.grad_sync is not None:
+            self.grad_sync(self.policy)
+        torch.nn.utils.clip_grad_norm_(self.policy.parameters(),
+                                       cfg.max_grad_norm)
+        self.opt.step()
+        return {"pi_loss": pi_loss.item(), "v_loss": v_loss.item(),
+                "entropy": ent.item(), "mean_reward": buf.reward.mean().item()}
+
+
+def train_impala(cfg, iterations=None, log_fn=print) -> dict:
+    """Async actor/learner loop (cfg is a trainer.TrainConfig)."""
+    from ..parallel.dist import (FlatGradAllreduce, broadcast_params,
+                                 init_from_env)
+    from ..trainer import make_env, make_policy, resolve_device
+    rank, world, local = init_from_env()
+    device = resolve_device(cfg)
+    if device.type == "cuda":
+        device = torch.device(f"cuda:{local}")
+    torch.manual_seed(cfg.seed * 31 + rank)
+    icfg = ImpalaConfig()
+    env = make_env(cfg, device, rank)
+    policy = make_policy(cfg, device)
+    broadcast_params(policy)
+    grad_sync = FlatGradAllreduce(policy) if world > 1 else None
+    learner = VtraceLearner(policy, icfg, grad_sync)
+    iters = iterations if iterations is not None else cfg.iterations
+    stats = {}
+
+    if device.type == "cuda":
+        from .rollout_gpu import GpuRolloutEngine
+        engine = GpuRolloutEngine(env, policy, cfg.rollout_T,
+                                  seed=cfg.seed * 101 + rank, num_bufs=2)
+        s_act = torch.cuda.Stream(device)
+        ev_collect = [torch.cuda.Event() for _ in range(2)]
+        ev_update = [torch.cuda.Event() for _ in range(2)]
+        ev_theta = [torch.cuda.Event() for _ in range(2)]
+        for e in ev_update + ev_theta:
+            e.record()  # buffers free; mirrors initialized at engine init
+        # Pipeline: the actor fills buf[i%2] with behavior mirror[i%2] on
+        # s_act while the learner trains on the other buffer on the default
+        # stream. After update j the learner refreshes mirror[(j)%2] (its
+        # next reader is rollout j+2) — staleness 1, no cross-stream races.
+        for it in range(iters + 1):
+            idx = it % 2
+            if it < iters:
+                with torch.cuda.stream(s_act):
+                    cur = torch.cuda.current_stream()
+                    cur.wait_event(ev_update[idx])  # buffer free
+                    cur.wait_event(ev_theta[idx])   # mirror refreshed
+                    engine.collect(idx, refresh=False)
+                    ev_collect[idx].record()
+            if it > 0:
+                prev = (it - 1) % 2
+                torch.cuda.current_stream().wait_event(ev_collect[prev])
+                stats = learner.update(engine.bufs[prev])
+                ev_update[prev].record()
+                if (it - 1) % icfg.behavior_sync == 0:
+                    engine.actors[prev].refresh()
+                ev_theta[prev].record()
+                if rank == 0 and log_fn and (it % 10 == 1 or it == iters):
+                    log_fn(f"[impala] update {it}: " +
+                           " ".join(f"{k}={v:.4f}" for k, v in stats.items()))
+        out = dict(stats)
+        out["episode_stats"] = env.episode_stats()
+        return out
+
+    # CPU fallback: explicit stale behavior copy, sequential
+    behavior = copy.deepcopy(policy)
+    collector = Collector(env, behavior, cfg.rollout_T, device,
+                          seed=cfg.seed * 101 + rank)
+    for it in range(iters):
+        if it % icfg.behavior_sync == 0:
+            behavior.load_state_dict(policy.state_dict())
+        buf = collector.collect()
+        stats = learner.update(buf)
+        if rank == 0 and log_fn and (it % 10 == 0 or it == iters - 1):
+            log_fn(f"[impala-cpu] iter {it}: " +
+                   " ".join(f"{k}={v:.4f}" for k, v in stats.items()))
+    out = dict(stats)
+    if hasattr(env, "venv"):
+        st = env.venv.stats
+        out["episode_stats"] = {
+            "episodes": st.count, "mean_return": st.mean_return(),
+            "mean_jct_us": st.mean_jct_us(), "completed": st.completed_sum,
+        }
+    return out

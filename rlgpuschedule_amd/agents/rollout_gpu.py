@@ -2,8 +2,12 @@
 
 Per rollout step exactly TWO kernel launches (actor_mlp, env_step), all
 outputs written straight into the rollout buffer slices (no copies). The
-whole T-step loop is captured in a hipGraph (torch.cuda.CUDAGraph) and
-replayed once per iteration — rollout cost becomes pure kernel time.
+T-step loop is captured in a hipGraph (one graph per buffer) and replayed
+once per iteration — rollout cost becomes pure kernel time.
+
+`num_bufs=2` double-buffers rollouts for the IMPALA async actor/learner
+(agents/impala.py): the actor stream fills buf[i] while the learner stream
+trains on buf[i^1]. Env state chains across rollouts via obs_last/mask_last.
 """
 from __future__ import annotations
 
@@ -15,36 +19,39 @@ from .rollout import RolloutBuffer
 
 class GpuRolloutEngine:
     def __init__(self, env, policy, T: int, seed: int = 0,
-                 use_graph: bool = True):
+                 use_graph: bool = True, num_bufs: int = 1):
         self.env = env
         self.policy = policy
         self.T = T
         self.device = torch.device("cuda")
-        self.actor = HipActorMLP(policy)
+        # one behavior-policy mirror per buffer: lets an async learner
+        # refresh mirror i (after its update) while a rollout on the other
+        # parity is still reading mirror i^1 (agents/impala.py)
+        self.actors = [HipActorMLP(policy) for _ in range(num_bufs)]
+        self.actor = self.actors[0]
         E = env.num_envs
         spec = env.spec
-        self.buf = RolloutBuffer(T, E, spec.obs_dim, spec.num_actions,
-                                 self.device)
+        self.bufs = [RolloutBuffer(T, E, spec.obs_dim, spec.num_actions,
+                                   self.device) for _ in range(num_bufs)]
+        self.buf = self.bufs[0]
         self.u_all = torch.empty(T, E, device=self.device)
         self.gen = torch.Generator(device=self.device)
         self.gen.manual_seed(seed)
         self.obs_last = torch.empty(E, spec.obs_dim, device=self.device)
         self.mask_last = torch.empty(E, spec.num_actions, dtype=torch.uint8,
                                      device=self.device)
-        self.rew_last = torch.empty(E, device=self.device)
-        self.done_last = torch.empty(E, dtype=torch.uint8, device=self.device)
         self.use_graph = use_graph
-        self._graph = None
-        # initial reset: env writes obs/mask into its default buffers;
-        # obs_last/mask_last always hold the NEXT rollout's start state
+        self._graphs: dict[int, torch.cuda.CUDAGraph] = {}
+        # initial reset: obs_last/mask_last always hold the NEXT rollout's
+        # start state
         env.reset()
         self.obs_last.copy_(env.b["obs"])
         self.mask_last.copy_(env.b["mask"])
 
-    def _rollout_body(self):
-        buf, env, T = self.buf, self.env, self.T
+    def _rollout_body(self, buf: RolloutBuffer, actor):
+        env, T = self.env, self.T
         for t in range(T):
-            self.actor.act(buf.obs[t], buf.mask[t], self.u_all[t],
+            actor.act(buf.obs[t], buf.mask[t], self.u_all[t],
                            buf.actions[t], buf.logp[t], buf.value[t])
             last = t == T - 1
             env.step(
@@ -54,31 +61,38 @@ class GpuRolloutEngine:
                 rew_out=buf.reward[t],
                 done_out=buf.done[t],
             )
-        self.actor.values(self.obs_last, buf.last_value, self.mask_last)
+        actor.values(self.obs_last, buf.last_value, self.mask_last)
 
-    def collect(self) -> RolloutBuffer:
-        self.actor.refresh()
+    def collect(self, idx: int = 0, refresh: bool = True) -> RolloutBuffer:
+        """Run one T-step rollout into bufs[idx] on the CURRENT stream.
+
+        The t=0 slices stay valid afterwards (the learner trains on them).
+        `refresh=False` keeps the behavior policy stale (IMPALA).
+        """
+        buf = self.bufs[idx]
+        actor = self.actors[idx]
+        if refresh:
+            actor.refresh()
         self.u_all.uniform_(generator=self.gen)
-        # rollout starts from the state the previous one ended in; the t=0
-        # slices must stay untouched afterwards (the learner reads them)
-        self.buf.obs[0].copy_(self.obs_last)
-        self.buf.mask[0].copy_(self.mask_last)
+        buf.obs[0].copy_(self.obs_last)
+        buf.mask[0].copy_(self.mask_last)
         if self.use_graph:
-            if self._graph is None:
-                # warm once on a side stream, then capture
+            if idx not in self._graphs:
                 s = torch.cuda.Stream()
                 s.wait_stream(torch.cuda.current_stream())
                 with torch.cuda.stream(s):
-                    self._rollout_body()
+                    self._rollout_body(buf, actor)   # warmup
                 torch.cuda.current_stream().wait_stream(s)
                 # warmup advanced the env: re-chain the replay's start state
-                self.buf.obs[0].copy_(self.obs_last)
-                self.buf.mask[0].copy_(self.mask_last)
+                buf.obs[0].copy_(self.obs_last)
+                buf.mask[0].copy_(self.mask_last)
                 torch.cuda.synchronize()
-                self._graph = torch.cuda.CUDAGraph()
-                with torch.cuda.graph(self._graph):
-                    self._rollout_body()
-            self._graph.replay()
+                g = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g):
+                    self._rollout_body(buf, actor)
+                self._graphs[idx] = g
+            self._graphs[idx].replay()
         else:
-            self._rollout_body()
-        return self.buf
+            self._rollout_body(buf, actor)
+        self.buf = buf
+        return buf

@@ -80,25 +80,44 @@ struct ActorArgs {
 #define ROWS 16      // env rows per workgroup
 #define HEADC 32     // head columns (A+1 <= 32)
 
+// K-permutation for float4 fragment reads: element k of a K-vector is
+// stored at kperm(k) = (k%4)*(K/4) + k/4 (bijective, K % 16 == 0). A lane
+// with MFMA k-subindex kk then reads its four k's for a 16-deep macro-step
+// as ONE aligned float4 at [kk*(K/4) + 4*k16]. Weights are packed to match
+// by models/hip_actor.py: wt[K,H] -> view(K/16,4(s),4(kk),H)
+// .permute(0,2,3,1) so a lane's B values for the macro-step are one float4
+// at (((k16*4+kk)*H + col)*4) — 16 consecutive lanes read 16 consecutive
+// float4s (fully coalesced 1 KiB per instruction).
+__device__ __forceinline__ int kperm(int k, int K) {
+  return (k & 3) * (K >> 2) + (k >> 2);
+}
+
 // One 64-column GEMM chunk for one wave: C[16 x 64] += A[16 x K] * B[K x 64].
-// A is an LDS tile with row stride lda; B global row-major [K][ldb].
-// Result written to LDS out (row stride ldo) through act_tanh epilogue.
+// A: LDS tile in kperm layout, row stride lda. B: packed global (see above),
+// logical row length ldb. Output written through bias(+tanh) epilogue into
+// LDS in kperm layout of the NEXT layer (perm_out) or plain (head).
 __device__ __forceinline__ void gemm_chunk_16x64(
     const float* __restrict__ a_lds, int lda, int K,
-    const float* __restrict__ b_glb, int ldb, int col0,
+    const float* __restrict__ b_pk, int ldb, int col0,
     const float* __restrict__ bias, bool do_tanh,
-    float* out_lds, int ldo, int lane) {
+    float* out_lds, int ldo, int lane, int perm_out, int Kout) {
   f32x4 acc[4] = {};
   const int ar = lane & 15;        // A row
-  const int kk = lane >> 4;        // A/B k sub-index (0..3)
+  const int kk = lane >> 4;        // k sub-index (0..3)
   const int bc = lane & 15;        // B col within 16-group
-  for (int k0 = 0; k0 < K; k0 += 4) {
-    float av = a_lds[ar * lda + k0 + kk];
-    const float* brow = b_glb + (int64_t)(k0 + kk) * ldb + col0;
+  const float* arow = a_lds + ar * lda + kk * (K >> 2);
+  const float* bbase = b_pk + (int64_t)4 * (kk * ldb + col0 + bc);
+  const int nk16 = K >> 4;
+  for (int k16 = 0; k16 < nk16; ++k16) {
+    f32x4 av = *(const f32x4*)(arow + 4 * k16);
+    const float* bk = bbase + (int64_t)16 * k16 * ldb;
     #pragma unroll
     for (int n = 0; n < 4; ++n) {
-      float bv = brow[n * 16 + bc];
-      acc[n] = __builtin_amdgcn_mfma_f32_16x16x4f32(av, bv, acc[n], 0, 0, 0);
+      f32x4 bv = *(const f32x4*)(bk + 64 * n);
+      #pragma unroll
+      for (int s = 0; s < 4; ++s)
+        acc[n] = __builtin_amdgcn_mfma_f32_16x16x4f32(av[s], bv[s], acc[n],
+                                                      0, 0, 0);
     }
   }
   // C/D map (16x16): col = lane&15, row = (lane>>4)*4 + i
@@ -111,7 +130,8 @@ __device__ __forceinline__ void gemm_chunk_16x64(
       int col = col0 + n * 16 + ccol;
       float vv = acc[n][i] + bias[col];
       if (do_tanh) vv = tanhf(vv);
-      out_lds[(crow + i) * ldo + col] = vv;
+      int oc = perm_out ? kperm(col, Kout) : col;
+      out_lds[(crow + i) * ldo + oc] = vv;
     }
   }
 }
@@ -122,46 +142,54 @@ __global__ __launch_bounds__(256) void actor_mlp_kernel(ActorArgs p) {
   const int w = tid >> 6;
   const int64_t row0 = (int64_t)blockIdx.x * ROWS;
   const int D = (int)p.D, H = (int)p.H, A = (int)p.A;
-  const int lda_o = D + 1, lda_h = H + 1;
+  const int Dp = (D + 15) & ~15;            // K-padded obs dim
+  // row strides: multiples of 4 floats (16B alignment for f32x4 reads) and
+  // != 0 mod 64 dwords so 16 rows hit 16 distinct bank groups
+  const int lda_o = Dp + 4, lda_h = H + 4;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  float* sobs = (float*)smem;                       // [ROWS][D+1]
-  float* h1 = sobs + ROWS * lda_o;                  // [ROWS][H+1]
-  float* h2 = h1 + ROWS * lda_h;                    // [ROWS][H+1]
+  float* sobs = (float*)smem;                       // [ROWS][Dp+4] kperm
+  float* h1 = sobs + ROWS * lda_o;                  // [ROWS][H+4] kperm
+  float* h2 = h1 + ROWS * lda_h;                    // [ROWS][H+4] kperm
   float* shead = h2 + ROWS * lda_h;                 // [ROWS][HEADC+1]
 
-  // stage obs tile (zero-pad rows past E and the k-pad column)
-  for (int ix = tid; ix < ROWS * lda_o; ix += 256) {
-    int r = ix / lda_o, d = ix % lda_o;
+  // stage obs tile into kperm layout (zero-pad rows past E and d >= D)
+  for (int ix = tid; ix < ROWS * Dp; ix += 256) {
+    int r = ix / Dp, d = ix % Dp;
     float vv = 0.0f;
     if (d < D && row0 + r < p.E) vv = p.obs[(row0 + r) * D + d];
-    sobs[r * lda_o + d] = vv;
+    sobs[r * lda_o + kperm(d, Dp)] = vv;
   }
   __syncthreads();
 
-  // layer 1: obs[16,D] @ wt1[D,H]
+  // layer 1: obs[16,Dp] @ wt1p[Dp,H]
   for (int c0 = w * 64; c0 < H; c0 += 256)
-    gemm_chunk_16x64(sobs, lda_o, D, p.wt1, H, c0, p.b1, true, h1, lda_h, lane);
-  // zero h1 pad column (k-loop over H reads index H? no: K=H, k0+kk < H+4?
-  // K=H is a multiple of 4 so k0+kk <= H-1; pad col never read as A)
+    gemm_chunk_16x64(sobs, lda_o, Dp, p.wt1, H, c0, p.b1, true,
+                     h1, lda_h, lane, 1, H);
   __syncthreads();
 
-  // layer 2: h1[16,H] @ wt2[H,H]
+  // layer 2: h1[16,H] @ wt2p[H,H]
   for (int c0 = w * 64; c0 < H; c0 += 256)
-    gemm_chunk_16x64(h1, lda_h, H, p.wt2, H, c0, p.b2, true, h2, lda_h, lane);
+    gemm_chunk_16x64(h1, lda_h, H, p.wt2, H, c0, p.b2, true,
+                     h2, lda_h, lane, 1, H);
   __syncthreads();
 
-  // head: h2[16,H] @ wth[H,32] — wave 0 only (tiny)
+  // head: h2[16,H] @ wthp[H,32] — wave 0 only (tiny)
   if (w == 0) {
     f32x4 acc[2] = {};
-    const int ar = lane & 15, kk = lane >> 4, bc = lane & 15;
-    for (int k0 = 0; k0 < H; k0 += 4) {
-      float av = h2[ar * lda_h + k0 + kk];
-      const float* brow = p.wth + (int64_t)(k0 + kk) * HEADC;
+    const int kk = lane >> 4, bc = lane & 15;
+    const float* arow = h2 + (lane & 15) * lda_h + kk * (H >> 2);
+    const float* bbase = p.wth + (int64_t)4 * (kk * HEADC + bc);
+    for (int k16 = 0; k16 < (H >> 4); ++k16) {
+      f32x4 av = *(const f32x4*)(arow + 4 * k16);
+      const float* bk = bbase + (int64_t)16 * k16 * HEADC;
       #pragma unroll
       for (int n = 0; n < 2; ++n) {
-        float bv = brow[n * 16 + bc];
-        acc[n] = __builtin_amdgcn_mfma_f32_16x16x4f32(av, bv, acc[n], 0, 0, 0);
+        f32x4 bv = *(const f32x4*)(bk + 64 * n);
+        #pragma unroll
+        for (int s = 0; s < 4; ++s)
+          acc[n] = __builtin_amdgcn_mfma_f32_16x16x4f32(av[s], bv[s], acc[n],
+                                                        0, 0, 0);
       }
     }
     const int crow = (lane >> 4) * 4, ccol = lane & 15;
@@ -227,7 +255,7 @@ extern "C" int rlsched_actor_mlp(const int64_t* ip, const uint64_t* bufs,
                                  void* stream) {
   ActorArgs p;
   p.E = ip[0]; p.D = ip[1]; p.H = ip[2]; p.A = ip[3]; p.value_only = ip[4];
-  if (p.D % 4 || p.H % 64 || p.A + 1 > HEADC) return -1;
+  if (p.H % 64 || p.A + 1 > HEADC) return -1;
   p.obs = (const float*)bufs[0];
   p.mask = (const uint8_t*)bufs[1];
   p.u = (const float*)bufs[2];
@@ -240,7 +268,8 @@ extern "C" int rlsched_actor_mlp(const int64_t* ip, const uint64_t* bufs,
   p.actions = (int64_t*)bufs[9];
   p.logp = (float*)bufs[10];
   p.value = (float*)bufs[11];
-  size_t lds = sizeof(float) * (ROWS * (p.D + 1) + 2 * ROWS * (p.H + 1)
+  const int Dp = ((int)p.D + 15) & ~15;
+  size_t lds = sizeof(float) * (ROWS * (Dp + 4) + 2 * ROWS * ((int)p.H + 4)
                                 + ROWS * (HEADC + 1));
   dim3 grid((unsigned)((p.E + ROWS - 1) / ROWS));
   hipLaunchKernelGGL(actor_mlp_kernel, grid, dim3(256), lds,
@@ -363,5 +392,69 @@ extern "C" int rlsched_ppo_loss(const int64_t* ip, const double* fp,
   if (grid > 2048) grid = 2048;
   hipLaunchKernelGGL(ppo_loss_kernel, dim3(grid), dim3(256), 0,
                      (hipStream_t)stream, a);
+  return (int)hipGetLastError();
+}
+
+// ------------------------------------------------------------- V-trace scan
+// IMPALA off-policy correction (Espeholt et al. 2018), same segmented
+// reverse-scan structure as gae_kernel: one thread per env, coalesced
+// [t][e] loads. Produces vs targets and policy-gradient advantages.
+namespace {
+
+struct VtraceArgs {
+  int64_t T, E;
+  float gamma, rho_bar, c_bar, lam;
+  const float* reward;      // [T*E]
+  const float* value;       // [T*E] V(x_t) under CURRENT params
+  const uint8_t* done;      // [T*E]
+  const float* last_value;  // [E]
+  const float* lp_cur;      // [T*E] log pi(a_t) current
+  const float* lp_beh;      // [T*E] log mu(a_t) behavior
+  float* vs;                // [T*E] out
+  float* pg_adv;            // [T*E] out
+};
+
+__global__ __launch_bounds__(256) void vtrace_kernel(VtraceArgs a) {
+  int64_t e = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (e >= a.E) return;
+  float acc = 0.0f;                  // vs_{t+1} - V_{t+1}
+  float next_v = a.last_value[e];    // V_{t+1}
+  float vs_next = a.last_value[e];   // vs_{t+1}
+  for (int64_t t = a.T - 1; t >= 0; --t) {
+    int64_t ix = t * a.E + e;
+    float nd = 1.0f - (float)a.done[ix];
+    float v = a.value[ix];
+    float r = a.reward[ix];
+    float ratio = __expf(a.lp_cur[ix] - a.lp_beh[ix]);
+    float rho = fminf(a.rho_bar, ratio);
+    float c = a.lam * fminf(a.c_bar, ratio);
+    float delta = rho * (r + a.gamma * next_v * nd - v);
+    acc = delta + a.gamma * c * nd * acc;
+    float vst = v + acc;
+    a.vs[ix] = vst;
+    a.pg_adv[ix] = rho * (r + a.gamma * vs_next * nd - v);
+    vs_next = vst;
+    next_v = v;
+  }
+}
+
+}  // namespace
+
+extern "C" int rlsched_vtrace(const int64_t* ip, const double* fp,
+                              const uint64_t* bufs, void* stream) {
+  VtraceArgs a;
+  a.T = ip[0]; a.E = ip[1];
+  a.gamma = (float)fp[0]; a.rho_bar = (float)fp[1];
+  a.c_bar = (float)fp[2]; a.lam = (float)fp[3];
+  a.reward = (const float*)bufs[0];
+  a.value = (const float*)bufs[1];
+  a.done = (const uint8_t*)bufs[2];
+  a.last_value = (const float*)bufs[3];
+  a.lp_cur = (const float*)bufs[4];
+  a.lp_beh = (const float*)bufs[5];
+  a.vs = (float*)bufs[6];
+  a.pg_adv = (float*)bufs[7];
+  dim3 grid((unsigned)((a.E + 255) / 256));
+  hipLaunchKernelGGL(vtrace_kernel, grid, dim3(256), 0, (hipStream_t)stream, a);
   return (int)hipGetLastError();
 }

@@ -19,28 +19,45 @@ class HipActorMLP:
             "fused actor kernel supports the 2-layer MLP flagship"
         self.policy = policy
         D, H, A = policy.obs_dim, policy.hidden, policy.num_actions
-        assert D % 4 == 0 and H % 64 == 0 and A + 1 <= HEADC
+        assert H % 64 == 0 and A + 1 <= HEADC
         dev = policy.pi.weight.device
         self.D, self.H, self.A = D, H, A
-        self.wt1 = torch.empty(D, H, device=dev)
+        self.Dp = (D + 15) & ~15
+        # packed layouts (see csrc/nn_kernels.hip kperm note):
+        # wt[K,H] -> view(K/16, 4(s), 4(kk), H).permute(0,2,3,1) contiguous
+        self.wt1 = torch.zeros(self.Dp // 16, 4, H, 4, device=dev)
         self.b1 = torch.empty(H, device=dev)
-        self.wt2 = torch.empty(H, H, device=dev)
+        self.wt2 = torch.empty(H // 16, 4, H, 4, device=dev)
         self.b2 = torch.empty(H, device=dev)
-        self.wth = torch.zeros(H, HEADC, device=dev)
+        self.wth = torch.zeros(H // 16, 4, HEADC, 4, device=dev)
         self.bh = torch.zeros(HEADC, device=dev)
         self.refresh()
+
+    @staticmethod
+    def _pack(wt: torch.Tensor) -> torch.Tensor:
+        """[K,H] -> [K/16, 4(kk), H, 4(s)] packed for f32x4 B-fragment reads."""
+        K, H = wt.shape
+        return wt.view(K // 16, 4, 4, H).permute(0, 2, 3, 1)
 
     @torch.no_grad()
     def refresh(self):
         p = self.policy
-        self.wt1.copy_(p.trunk[0].weight.t())
+        D, H, A = self.D, self.H, self.A
+        if not hasattr(self, "_wt1_full"):
+            self._wt1_full = self.wt1.new_zeros(self.Dp, H)
+            self._wth_full = self.wth.new_zeros(H, HEADC)
+        wt1_full = self._wt1_full
+        wt1_full[:D].copy_(p.trunk[0].weight.t())
+        self.wt1.copy_(self._pack(wt1_full))
         self.b1.copy_(p.trunk[0].bias)
-        self.wt2.copy_(p.trunk[1].weight.t())
+        self.wt2.copy_(self._pack(p.trunk[1].weight.t()))
         self.b2.copy_(p.trunk[1].bias)
-        self.wth[:, : self.A].copy_(p.pi.weight.t())
-        self.wth[:, self.A].copy_(p.v.weight[0])
-        self.bh[: self.A].copy_(p.pi.bias)
-        self.bh[self.A] = p.v.bias[0]
+        wth_full = self._wth_full
+        wth_full[:, :A].copy_(p.pi.weight.t())
+        wth_full[:, A].copy_(p.v.weight[0])
+        self.wth.copy_(self._pack(wth_full))
+        self.bh[:A].copy_(p.pi.bias)
+        self.bh[A] = p.v.bias[0]
 
     def act(self, obs, mask, u, actions, logp, value):
         """Sample actions in-place into the given out tensors (one launch)."""

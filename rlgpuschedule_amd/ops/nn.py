@@ -57,7 +57,7 @@ def actor_mlp(obs, mask, u, wt1, b1, wt2, b2, wth, bh, actions, logp, value,
               value_only: bool = False):
     """Fused MLP policy forward + masked inverse-CDF sampling (one launch)."""
     E, D = obs.shape
-    H = wt1.shape[1]
+    H = wt2.shape[2]  # wt2 packed [H/16, 4, H, 4]
     A = mask.shape[1]
     ip = np.array([E, D, H, A, 1 if value_only else 0], np.int64)
     bufs = _ptrs(obs, mask, u, wt1, b1, wt2, b2, wth, bh, actions, logp, value)
@@ -94,3 +94,30 @@ def ppo_loss_fused(logits, value, actions, old_logp, old_value, adv, ret,
         bufs.ctypes.data_as(ctypes.c_void_p), ctypes.c_void_p(_stream(logits)))
     if rc != 0:
         raise RuntimeError(f"rlsched_ppo_loss failed: {rc}")
+
+
+def _load_vtrace():
+    lib = load()
+    if not hasattr(lib, "_vtrace_ready"):
+        lib.rlsched_vtrace.restype = ctypes.c_int
+        lib.rlsched_vtrace.argtypes = [ctypes.c_void_p] * 4
+        lib._vtrace_ready = True
+    return lib
+
+
+def vtrace_scan(reward, value, done, last_value, lp_cur, lp_beh,
+                gamma, rho_bar=1.0, c_bar=1.0, lam=1.0,
+                vs=None, pg_adv=None):
+    """HIP V-trace reverse scan. All [T,E] f32 (done u8). Returns (vs, pg_adv)."""
+    T, E = reward.shape
+    vs = torch.empty_like(reward) if vs is None else vs
+    pg_adv = torch.empty_like(reward) if pg_adv is None else pg_adv
+    ip = np.array([T, E], np.int64)
+    fp = np.array([gamma, rho_bar, c_bar, lam], np.float64)
+    bufs = _ptrs(reward, value, done, last_value, lp_cur, lp_beh, vs, pg_adv)
+    rc = _load_vtrace().rlsched_vtrace(
+        ip.ctypes.data_as(ctypes.c_void_p), fp.ctypes.data_as(ctypes.c_void_p),
+        bufs.ctypes.data_as(ctypes.c_void_p), ctypes.c_void_p(_stream(reward)))
+    if rc != 0:
+        raise RuntimeError(f"rlsched_vtrace failed: {rc}")
+    return vs, pg_adv
