@@ -42,9 +42,12 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=12)
     ap.add_argument("--warmup", type=int, default=3)
-    ap.add_argument("--envs", type=int, default=1024, help="envs per GPU")
+    ap.add_argument("--envs", type=int, default=4096, help="envs per GPU")
     ap.add_argument("--rollout", type=int, default=64)
     ap.add_argument("--hidden", type=int, default=256)
+    ap.add_argument("--learner", choices=["hip", "fused"], default="hip",
+                    help="hip: fully hand-written update kernels; fused: "
+                         "torch GEMMs + fused HIP loss")
     args = ap.parse_args()
 
     rank, world, local = pdist.init_from_env()
@@ -71,8 +74,10 @@ def main():
     cfg = PPOConfig(epochs=1, minibatches=8)
     if has_gpu:
         from rlgpuschedule_amd.agents.ppo_gpu import FusedPPOLearner
+        from rlgpuschedule_amd.agents.ppo_hip import HipPPOLearner
         from rlgpuschedule_amd.agents.rollout_gpu import GpuRolloutEngine
-        learner = FusedPPOLearner(policy, cfg, grad_sync)
+        lcls = HipPPOLearner if args.learner == "hip" else FusedPPOLearner
+        learner = lcls(policy, cfg, grad_sync)
         collector = GpuRolloutEngine(env, policy, T, seed=555 + rank)
     else:
         learner = PPOLearner(policy, cfg, grad_sync)
@@ -125,6 +130,7 @@ def main():
                 "seq_len": T,
                 "envs_per_gpu": E,
                 "ppo": "1 epoch x 8 minibatches, clip 0.2, GAE(0.99, 0.95)",
+                "update_path": args.learner,
                 "parallelism": f"dp{n_gpus}",
             },
         }

@@ -21,12 +21,7 @@
 //
 // Numerics tests: tests/test_nn_kernels.py vs torch fp32 reference.
 
-#include <hip/hip_runtime.h>
-#include <stdint.h>
-
-#define WAVE 64
-
-typedef float f32x4 __attribute__((ext_vector_type(4)));
+#include "nn_common.h"
 
 namespace {
 
@@ -88,10 +83,6 @@ struct ActorArgs {
 // .permute(0,2,3,1) so a lane's B values for the macro-step are one float4
 // at (((k16*4+kk)*H + col)*4) — 16 consecutive lanes read 16 consecutive
 // float4s (fully coalesced 1 KiB per instruction).
-__device__ __forceinline__ int kperm(int k, int K) {
-  return (k & 3) * (K >> 2) + (k >> 2);
-}
-
 // One 64-column GEMM chunk for one wave: C[16 x 64] += A[16 x K] * B[K x 64].
 // A: LDS tile in kperm layout, row stride lda. B: packed global (see above),
 // logical row length ldb. Output written through bias(+tanh) epilogue into

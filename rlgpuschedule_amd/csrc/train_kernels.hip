@@ -1,0 +1,425 @@
+// Hand-written CDNA4 training kernels for the PPO/IMPALA update path:
+// MLP forward (activation-saving), masked PPO loss head (fwd+bwd), backward
+// dX GEMMs with fused tanh', backward dW GEMMs (LDS-tiled, split over
+// batch chunks with atomic accumulation), and bias column sums.
+// All GEMMs are MFMA f32 16x16x4 (exact f32) with the kperm/f32x4 fragment
+// scheme of nn_common.h. Replaces torch/hipBLASLt in the update so the
+// whole PPO iteration is hand-written HIP (BASELINE.json north star).
+//
+// Gradient-check tests: tests/test_train_kernels.py vs torch autograd.
+
+#include "nn_common.h"
+
+#define ROWS 16
+#define HEADC 32
+
+namespace {
+
+// --------------------------------------------------- training MLP forward
+// Saves h1,h2 (plain [B,H]) and head [B,HEADC] for the backward pass.
+struct FwdTrainArgs {
+  int64_t B, D, H;
+  const float* obs;   // [B,D]
+  const float* wt1;   // packed [Dp,H]
+  const float* b1;
+  const float* wt2;   // packed [H,H]
+  const float* b2;
+  const float* wth;   // packed [H,HEADC]
+  const float* bh;
+  float* h1;          // [B,H] out (plain)
+  float* h2;          // [B,H] out
+  float* head;        // [B,HEADC] out
+};
+
+// gemm chunk with epilogue writing BOTH kperm LDS (next layer's A) and a
+// plain global activation row block.
+__device__ __forceinline__ void chunk_fwd_train(
+    const float* a_lds, int lda, int K, const float* b_pk, int ldb, int col0,
+    const float* bias, float* out_lds, int ldo, float* gout, int64_t row0,
+    int64_t B, int ldg, int lane) {
+  f32x4 acc[4] = {};
+  mfma_core_16x64(a_lds, lda, K, b_pk, ldb, col0, lane, acc);
+  const int crow = (lane >> 4) * 4, ccol = lane & 15;
+  #pragma unroll
+  for (int n = 0; n < 4; ++n)
+    #pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      int col = col0 + n * 16 + ccol;
+      float vv = tanhf(acc[n][i] + bias[col]);
+      out_lds[(crow + i) * ldo + kperm(col, ldg)] = vv;
+      int64_t row = row0 + crow + i;
+      if (row < B) gout[row * ldg + col] = vv;
+    }
+}
+
+__global__ __launch_bounds__(256) void fwd_train_kernel(FwdTrainArgs p) {
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int w = tid >> 6;
+  const int64_t row0 = (int64_t)blockIdx.x * ROWS;
+  const int D = (int)p.D, H = (int)p.H;
+  const int Dp = (D + 15) & ~15;
+  const int lda_o = Dp + 4, lda_h = H + 4;
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* sobs = (float*)smem;
+  float* sh1 = sobs + ROWS * lda_o;
+  float* sh2 = sh1 + ROWS * lda_h;
+  for (int ix = tid; ix < ROWS * Dp; ix += 256) {
+    int r = ix / Dp, d = ix % Dp;
+    float vv = 0.0f;
+    if (d < D && row0 + r < p.B) vv = p.obs[(row0 + r) * D + d];
+    sobs[r * lda_o + kperm(d, Dp)] = vv;
+  }
+  __syncthreads();
+  for (int c0 = w * 64; c0 < H; c0 += 256)
+    chunk_fwd_train(sobs, lda_o, Dp, p.wt1, H, c0, p.b1, sh1, lda_h,
+                    p.h1, row0, p.B, H, lane);
+  __syncthreads();
+  for (int c0 = w * 64; c0 < H; c0 += 256)
+    chunk_fwd_train(sh1, lda_h, H, p.wt2, H, c0, p.b2, sh2, lda_h,
+                    p.h2, row0, p.B, H, lane);
+  __syncthreads();
+  if (w == 0) {
+    f32x4 acc[2] = {};
+    const int kk = lane >> 4, bc = lane & 15;
+    const float* arow = sh2 + (lane & 15) * lda_h + kk * (H >> 2);
+    const float* bbase = p.wth + (int64_t)4 * (kk * HEADC + bc);
+    for (int k16 = 0; k16 < (H >> 4); ++k16) {
+      f32x4 av = *(const f32x4*)(arow + 4 * k16);
+      const float* bk = bbase + (int64_t)16 * k16 * HEADC;
+      #pragma unroll
+      for (int n = 0; n < 2; ++n) {
+        f32x4 bv = *(const f32x4*)(bk + 64 * n);
+        #pragma unroll
+        for (int s = 0; s < 4; ++s)
+          acc[n] = __builtin_amdgcn_mfma_f32_16x16x4f32(av[s], bv[s], acc[n],
+                                                        0, 0, 0);
+      }
+    }
+    const int crow = (lane >> 4) * 4, ccol = lane & 15;
+    #pragma unroll
+    for (int n = 0; n < 2; ++n)
+      #pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        int col = n * 16 + ccol;
+        int64_t row = row0 + crow + i;
+        if (row < p.B) p.head[row * HEADC + col] = acc[n][i] + p.bh[col];
+      }
+  }
+}
+
+// -------------------------------------------- masked PPO loss on head rows
+// Like ppo_loss_kernel (nn_kernels.hip) but reads the raw head [B,HEADC]
+// (logits cols [0,A), value col A), applies the action mask itself
+// (invalid actions excluded from the softmax), and writes the fused
+// gradient dhead [B,HEADC] (zeros for masked/pad columns).
+struct PpoHeadArgs {
+  int64_t B, A;
+  float clip, vf_coef, ent_coef, inv_n;
+  const float* head;       // [B,HEADC]
+  const uint8_t* mask;     // [B,A]
+  const int64_t* actions;  // [B]
+  const float* old_logp;
+  const float* old_value;
+  const float* adv;
+  const float* ret;
+  float* dhead;            // [B,HEADC] out
+  float* stats;            // [4] sums
+};
+
+__global__ __launch_bounds__(256) void ppo_head_kernel(PpoHeadArgs a) {
+  const int64_t i0 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int A = (int)a.A;
+  float s_pi = 0.f, s_v = 0.f, s_ent = 0.f, s_cf = 0.f;
+  for (int64_t i = i0; i < a.B; i += (int64_t)gridDim.x * blockDim.x) {
+    const float* lg = a.head + i * HEADC;
+    const uint8_t* mk = a.mask + i * A;
+    float* dlg = a.dhead + i * HEADC;
+    const int act = (int)a.actions[i];
+    float mx = -3.0e38f;
+    for (int j = 0; j < A; ++j)
+      if (mk[j]) mx = fmaxf(mx, lg[j]);
+    float sum = 0.f;
+    for (int j = 0; j < A; ++j)
+      if (mk[j]) sum += __expf(lg[j] - mx);
+    const float lsum = __logf(sum);
+    const float logp_a = (lg[act] - mx) - lsum;
+    float H = 0.f;
+    for (int j = 0; j < A; ++j)
+      if (mk[j]) {
+        float lpj = (lg[j] - mx) - lsum;
+        H -= __expf(lpj) * lpj;
+      }
+    const float advi = a.adv[i];
+    const float ratio = __expf(logp_a - a.old_logp[i]);
+    const float rcl = fminf(fmaxf(ratio, 1.f - a.clip), 1.f + a.clip);
+    const float s1 = ratio * advi, s2 = rcl * advi;
+    const float active = (s1 <= s2) ? 1.f : 0.f;
+    s_pi += -fminf(s1, s2);
+    s_ent += H;
+    s_cf += (fabsf(ratio - 1.f) > a.clip) ? 1.f : 0.f;
+    const float g_logp = -advi * ratio * active * a.inv_n;
+    for (int j = 0; j < A; ++j) {
+      float g = 0.f;
+      if (mk[j]) {
+        float lpj = (lg[j] - mx) - lsum;
+        float pj = __expf(lpj);
+        g = g_logp * (((j == act) ? 1.f : 0.f) - pj)
+          + a.ent_coef * pj * (lpj + H) * a.inv_n;
+      }
+      dlg[j] = g;
+    }
+    for (int j = A + 1; j < HEADC; ++j) dlg[j] = 0.f;
+    const float v = lg[A], ov = a.old_value[i], rt = a.ret[i];
+    const float vcl = ov + fminf(fmaxf(v - ov, -a.clip), a.clip);
+    const float e1 = v - rt, e2 = vcl - rt;
+    const float q1 = e1 * e1, q2 = e2 * e2;
+    s_v += 0.5f * fmaxf(q1, q2);
+    float dv;
+    if (q1 >= q2) dv = e1;
+    else dv = (fabsf(v - ov) < a.clip) ? e2 : 0.f;
+    dlg[A] = a.vf_coef * dv * a.inv_n;
+  }
+  __shared__ float red[4][256 / WAVE];
+  float vals[4] = {s_pi, s_v, s_ent, s_cf};
+  const int lane = threadIdx.x & (WAVE - 1), w = threadIdx.x >> 6;
+  #pragma unroll
+  for (int k = 0; k < 4; ++k) {
+    float x = vals[k];
+    #pragma unroll
+    for (int off = 32; off > 0; off >>= 1) x += __shfl_xor(x, off, WAVE);
+    if (lane == 0) red[k][w] = x;
+  }
+  __syncthreads();
+  if (threadIdx.x < 4) {
+    float x = 0.f;
+    for (int q = 0; q < 256 / WAVE; ++q) x += red[threadIdx.x][q];
+    atomicAdd(&a.stats[threadIdx.x], x);
+  }
+}
+
+// ------------------------------------- backward dX = (dY @ W^T) * tanh'(h)
+// dY [B,K] (K = HEADC or H), W^T packed [K,N], h [B,N] saved activation.
+struct BwdDxArgs {
+  int64_t B, K, N;
+  const float* dY;    // [B,K]
+  const float* wtp;   // packed [K,N]
+  const float* h;     // [B,N]
+  float* dX;          // [B,N] out
+};
+
+__global__ __launch_bounds__(256) void bwd_dx_kernel(BwdDxArgs p) {
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int w = tid >> 6;
+  const int64_t row0 = (int64_t)blockIdx.x * ROWS;
+  const int K = (int)p.K, N = (int)p.N;
+  const int lda = K + 4;
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* sdy = (float*)smem;    // [ROWS][K+4] kperm
+  for (int ix = tid; ix < ROWS * K; ix += 256) {
+    int r = ix / K, k = ix % K;
+    float vv = 0.0f;
+    if (row0 + r < p.B) vv = p.dY[(row0 + r) * K + k];
+    sdy[r * lda + kperm(k, K)] = vv;
+  }
+  __syncthreads();
+  for (int c0 = w * 64; c0 < N; c0 += 256) {
+    f32x4 acc[4] = {};
+    mfma_core_16x64(sdy, lda, K, p.wtp, N, c0, lane, acc);
+    const int crow = (lane >> 4) * 4, ccol = lane & 15;
+    #pragma unroll
+    for (int n = 0; n < 4; ++n)
+      #pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        int col = c0 + n * 16 + ccol;
+        int64_t row = row0 + crow + i;
+        if (row < p.B) {
+          float hv = p.h[row * N + col];
+          p.dX[row * N + col] = acc[n][i] * (1.0f - hv * hv);
+        }
+      }
+  }
+}
+
+// ----------------------------------------- backward dW += X^T @ dY over B
+// X [B,M], dY [B,N] -> dW [M,N] (f32 atomicAdd; caller zeroes dW).
+// Block: 64(m) x 64(n) C-tile x BC-deep batch chunk; both operands staged
+// transposed into kperm LDS tiles ([64 c-dim][64 kperm] padded).
+struct BwdDwArgs {
+  int64_t B, M, N, BC;   // BC: batch rows per block (multiple of 64)
+  const float* X;        // [B,M]
+  const float* dY;       // [B,N]
+  float* dW;             // [M,N] accumulated
+};
+
+#define DW_LD 68   // 64 + 4 pad (stride % 64 == 4 -> distinct bank groups)
+
+__global__ __launch_bounds__(256) void bwd_dw_kernel(BwdDwArgs p) {
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int w = tid >> 6;
+  const int M = (int)p.M, N = (int)p.N;
+  const int mtiles = (M + 63) >> 6, ntiles = (N + 63) >> 6;
+  int bid = blockIdx.x;
+  const int m0 = (bid % mtiles) * 64; bid /= mtiles;
+  const int n0 = (bid % ntiles) * 64; bid /= ntiles;
+  const int64_t b0 = (int64_t)bid * p.BC;
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* xt = (float*)smem;          // [64 m][DW_LD] kperm over k
+  float* yt = xt + 64 * DW_LD;       // [64 n][DW_LD]
+  f32x4 acc[4] = {};
+  const int wm = w * 16;             // wave's 16 m-rows within the tile
+  const int ar = lane & 15, kk = lane >> 4, bc = lane & 15;
+  for (int kc = 0; kc < (int)p.BC; kc += 64) {
+    // stage X^T and dY^T chunks: global row b (=k), LDS row = feature col,
+    // k stored kperm so a lane's 4 k's per macro-step are one float4
+    for (int ix = tid; ix < 64 * 64; ix += 256) {
+      int k = ix >> 6, j = ix & 63;   // k: batch row in chunk, j: feature
+      int64_t b = b0 + kc + k;
+      int kp = kperm(k, 64);
+      float xv = 0.f, yv = 0.f;
+      if (b < p.B) {
+        if (m0 + j < M) xv = p.X[b * M + m0 + j];
+        if (n0 + j < N) yv = p.dY[b * N + n0 + j];
+      }
+      xt[j * DW_LD + kp] = xv;
+      yt[j * DW_LD + kp] = yv;
+    }
+    __syncthreads();
+    // wave computes C[wm..wm+15][0..63] over this 64-deep k chunk; both
+    // fragments are float4 LDS reads (kperm: k = 16*k16 + 4s + kk ->
+    // position kk*16 + 4*k16 + s)
+    const float* arow = xt + (wm + ar) * DW_LD + kk * 16;
+    const float* brows = yt + bc * DW_LD + kk * 16;
+    #pragma unroll
+    for (int k16 = 0; k16 < 4; ++k16) {
+      f32x4 av = *(const f32x4*)(arow + 4 * k16);
+      #pragma unroll
+      for (int n = 0; n < 4; ++n) {
+        f32x4 bv = *(const f32x4*)(brows + n * 16 * DW_LD + 4 * k16);
+        #pragma unroll
+        for (int s = 0; s < 4; ++s)
+          acc[n] = __builtin_amdgcn_mfma_f32_16x16x4f32(av[s], bv[s], acc[n],
+                                                        0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+  const int crow = (lane >> 4) * 4, ccol = lane & 15;
+  #pragma unroll
+  for (int n = 0; n < 4; ++n)
+    #pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      int m = m0 + wm + crow + i;
+      int nn = n0 + n * 16 + ccol;
+      if (m < M && nn < N) atomicAdd(&p.dW[(int64_t)m * N + nn], acc[n][i]);
+    }
+}
+
+// ------------------------------------------------------- bias column sums
+// db[N] = sum_b dY[b][n] (mean scaling already inside dY via the loss).
+struct ColSumArgs {
+  int64_t B, N;
+  const float* dY;
+  float* db;
+};
+
+__global__ __launch_bounds__(256) void colsum_kernel(ColSumArgs p) {
+  int64_t n = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (n >= p.N) return;
+  float s = 0.f;
+  for (int64_t b = 0; b < p.B; ++b) s += p.dY[b * p.N + n];
+  p.db[n] = s;
+}
+
+}  // namespace
+
+extern "C" int rlsched_fwd_train(const int64_t* ip, const uint64_t* bufs,
+                                 void* stream) {
+  FwdTrainArgs p;
+  p.B = ip[0]; p.D = ip[1]; p.H = ip[2];
+  if (p.H % 64) return -1;
+  p.obs = (const float*)bufs[0];
+  p.wt1 = (const float*)bufs[1];
+  p.b1 = (const float*)bufs[2];
+  p.wt2 = (const float*)bufs[3];
+  p.b2 = (const float*)bufs[4];
+  p.wth = (const float*)bufs[5];
+  p.bh = (const float*)bufs[6];
+  p.h1 = (float*)bufs[7];
+  p.h2 = (float*)bufs[8];
+  p.head = (float*)bufs[9];
+  const int Dp = ((int)p.D + 15) & ~15;
+  size_t lds = sizeof(float) * (ROWS * (Dp + 4) + 2 * ROWS * ((int)p.H + 4));
+  dim3 grid((unsigned)((p.B + ROWS - 1) / ROWS));
+  hipLaunchKernelGGL(fwd_train_kernel, grid, dim3(256), lds,
+                     (hipStream_t)stream, p);
+  return (int)hipGetLastError();
+}
+
+extern "C" int rlsched_ppo_head(const int64_t* ip, const double* fp,
+                                const uint64_t* bufs, void* stream) {
+  PpoHeadArgs a;
+  a.B = ip[0]; a.A = ip[1];
+  a.clip = (float)fp[0]; a.vf_coef = (float)fp[1];
+  a.ent_coef = (float)fp[2]; a.inv_n = (float)fp[3];
+  a.head = (const float*)bufs[0];
+  a.mask = (const uint8_t*)bufs[1];
+  a.actions = (const int64_t*)bufs[2];
+  a.old_logp = (const float*)bufs[3];
+  a.old_value = (const float*)bufs[4];
+  a.adv = (const float*)bufs[5];
+  a.ret = (const float*)bufs[6];
+  a.dhead = (float*)bufs[7];
+  a.stats = (float*)bufs[8];
+  int grid = (int)((a.B + 255) / 256);
+  if (grid > 2048) grid = 2048;
+  hipLaunchKernelGGL(ppo_head_kernel, dim3(grid), dim3(256), 0,
+                     (hipStream_t)stream, a);
+  return (int)hipGetLastError();
+}
+
+extern "C" int rlsched_bwd_dx(const int64_t* ip, const uint64_t* bufs,
+                              void* stream) {
+  BwdDxArgs p;
+  p.B = ip[0]; p.K = ip[1]; p.N = ip[2];
+  if (p.K % 16 || p.N % 64) return -1;
+  p.dY = (const float*)bufs[0];
+  p.wtp = (const float*)bufs[1];
+  p.h = (const float*)bufs[2];
+  p.dX = (float*)bufs[3];
+  size_t lds = sizeof(float) * ROWS * ((int)p.K + 4);
+  dim3 grid((unsigned)((p.B + ROWS - 1) / ROWS));
+  hipLaunchKernelGGL(bwd_dx_kernel, grid, dim3(256), lds,
+                     (hipStream_t)stream, p);
+  return (int)hipGetLastError();
+}
+
+extern "C" int rlsched_bwd_dw(const int64_t* ip, const uint64_t* bufs,
+                              void* stream) {
+  BwdDwArgs p;
+  p.B = ip[0]; p.M = ip[1]; p.N = ip[2]; p.BC = ip[3];
+  if (p.BC % 64) return -1;  // batch-row guards handle B % 64 != 0
+  p.X = (const float*)bufs[0];
+  p.dY = (const float*)bufs[1];
+  p.dW = (float*)bufs[2];
+  const int mtiles = ((int)p.M + 63) >> 6, ntiles = ((int)p.N + 63) >> 6;
+  const int chunks = (int)((p.B + p.BC - 1) / p.BC);
+  size_t lds = sizeof(float) * 2 * 64 * DW_LD;
+  dim3 grid((unsigned)(mtiles * ntiles * chunks));
+  hipLaunchKernelGGL(bwd_dw_kernel, grid, dim3(256), lds,
+                     (hipStream_t)stream, p);
+  return (int)hipGetLastError();
+}
+
+extern "C" int rlsched_colsum(const int64_t* ip, const uint64_t* bufs,
+                              void* stream) {
+  ColSumArgs p;
+  p.B = ip[0]; p.N = ip[1];
+  p.dY = (const float*)bufs[0];
+  p.db = (float*)bufs[1];
+  dim3 grid((unsigned)((p.N + 255) / 256));
+  hipLaunchKernelGGL(colsum_kernel, grid, dim3(256), 0, (hipStream_t)stream, p);
+  return (int)hipGetLastError();
+}

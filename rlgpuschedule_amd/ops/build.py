@@ -20,6 +20,7 @@ ARCH = os.environ.get("RLS_OFFLOAD_ARCH", "gfx950")
 _LIBS = {
     "librlsched_env.so": ["env_step.hip"],
     "librlsched_nn.so": ["nn_kernels.hip"],
+    "librlsched_train.so": ["train_kernels.hip"],
 }
 
 
