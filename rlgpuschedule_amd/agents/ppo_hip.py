@@ -112,7 +112,7 @@ class HipPPOLearner:
                          cfg.ent_coef, s["dhead"], s["stats"])
                 bwd_dx(s["dhead"], mi.whT, s["h2"], s["dh2"])
                 bwd_dx(s["dh2"], mi.w2T, s["h1"], s["dh1"])
-                for k in ("dW1", "dW2", "dWh"):
+                for k in ("dW1", "dW2", "dWh", "db1", "db2", "dbh"):
                     s[k].zero_()
                 bwd_dw(s["h2"], s["dhead"], s["dWh"])
                 bwd_dw(s["h1"], s["dh2"], s["dW2"])

@@ -13,7 +13,7 @@ from __future__ import annotations
 
 import torch
 
-from ..models.hip_actor import HipActorMLP
+from ..models.hip_actor import HipActorGRU, HipActorMLP
 from .rollout import RolloutBuffer
 
 
@@ -27,12 +27,20 @@ class GpuRolloutEngine:
         # one behavior-policy mirror per buffer: lets an async learner
         # refresh mirror i (after its update) while a rollout on the other
         # parity is still reading mirror i^1 (agents/impala.py)
-        self.actors = [HipActorMLP(policy) for _ in range(num_bufs)]
+        acls = HipActorGRU if policy.is_recurrent else HipActorMLP
+        self.actors = [acls(policy) for _ in range(num_bufs)]
         self.actor = self.actors[0]
         E = env.num_envs
         spec = env.spec
+        hid = policy.hidden if policy.is_recurrent else 0
         self.bufs = [RolloutBuffer(T, E, spec.obs_dim, spec.num_actions,
-                                   self.device) for _ in range(num_bufs)]
+                                   self.device, hidden=hid)
+                     for _ in range(num_bufs)]
+        if policy.is_recurrent:
+            self.hx = torch.zeros(E, policy.hidden, device=self.device)
+            self.hx_scratch = torch.empty_like(self.hx)
+            self.done_last = torch.zeros(E, dtype=torch.uint8,
+                                         device=self.device)
         self.buf = self.bufs[0]
         self.u_all = torch.empty(T, E, device=self.device)
         self.gen = torch.Generator(device=self.device)
@@ -50,9 +58,20 @@ class GpuRolloutEngine:
 
     def _rollout_body(self, buf: RolloutBuffer, actor):
         env, T = self.env, self.T
+        recurrent = self.policy.is_recurrent
+        if recurrent:
+            # learner replays from h0 (already reset where done_last)
+            buf.h0.copy_(self.hx
+                         * (1.0 - self.done_last.float()).unsqueeze(-1))
         for t in range(T):
-            actor.act(buf.obs[t], buf.mask[t], self.u_all[t],
-                           buf.actions[t], buf.logp[t], buf.value[t])
+            if recurrent:
+                dprev = self.done_last if t == 0 else buf.done[t - 1]
+                actor.act(buf.obs[t], buf.mask[t], self.u_all[t], dprev,
+                          self.hx, self.hx, buf.actions[t], buf.logp[t],
+                          buf.value[t])
+            else:
+                actor.act(buf.obs[t], buf.mask[t], self.u_all[t],
+                          buf.actions[t], buf.logp[t], buf.value[t])
             last = t == T - 1
             env.step(
                 buf.actions[t],
@@ -61,7 +80,12 @@ class GpuRolloutEngine:
                 rew_out=buf.reward[t],
                 done_out=buf.done[t],
             )
-        actor.values(self.obs_last, buf.last_value, self.mask_last)
+        if recurrent:
+            self.done_last.copy_(buf.done[T - 1])
+            actor.values(self.obs_last, self.mask_last, self.done_last,
+                         self.hx, self.hx_scratch, buf.last_value)
+        else:
+            actor.values(self.obs_last, buf.last_value, self.mask_last)
 
     def collect(self, idx: int = 0, refresh: bool = True) -> RolloutBuffer:
         """Run one T-step rollout into bufs[idx] on the CURRENT stream.

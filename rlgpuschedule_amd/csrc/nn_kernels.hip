@@ -449,3 +449,213 @@ extern "C" int rlsched_vtrace(const int64_t* ip, const double* fp,
   hipLaunchKernelGGL(vtrace_kernel, grid, dim3(256), 0, (hipStream_t)stream, a);
   return (int)hipGetLastError();
 }
+
+// ---------------------------------------------------------- fused GRU actor
+// One launch per rollout step for the recurrent policy (GRUPolicy):
+//   x  = tanh(obs @ Win + bin)
+//   r  = sigmoid(x@Wir + bir + h@Whr + bhr)        (torch GRUCell order
+//   z  = sigmoid(x@Wiz + biz + h@Whz + bhz)         r, z, n)
+//   n  = tanh(x@Win2 + bin2 + r * (h@Whn + bhn))
+//   h' = (1-z)*n + z*h ;  head = h' @ Wht + bh ; masked sample
+// Gate GEMMs run chunk-wise (64 cols x 6 gate accumulators in registers),
+// so the 3H pre-activations are never materialized. h is zeroed for envs
+// whose previous transition ended an episode (done_prev).
+namespace {
+
+struct GruActorArgs {
+  int64_t E, D, H, A;
+  int64_t value_only;
+  const float* obs;        // [E,D]
+  const uint8_t* mask;     // [E,A]
+  const float* u;          // [E]
+  const uint8_t* done_prev;  // [E] or null: zero h where set
+  const float* hx;         // [E,H] in
+  float* hx_out;           // [E,H] out (may alias hx)
+  const float* wtin;       // packed [Dp,H]
+  const float* bin;        // [H]
+  const float* wih;        // packed [H,3H] (weight_ih.t())
+  const float* bih;        // [3H]
+  const float* whh;        // packed [H,3H]
+  const float* bhh;        // [3H]
+  const float* wth;        // packed [H,HEADC]
+  const float* bh;         // [HEADC]
+  int64_t* actions;
+  float* logp;
+  float* value;
+};
+
+__device__ __forceinline__ float sigmoidf_(float x) {
+  return 1.0f / (1.0f + __expf(-x));
+}
+
+__global__ __launch_bounds__(256) void gru_actor_kernel(GruActorArgs p) {
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int w = tid >> 6;
+  const int64_t row0 = (int64_t)blockIdx.x * ROWS;
+  const int D = (int)p.D, H = (int)p.H, A = (int)p.A;
+  const int Dp = (D + 15) & ~15;
+  const int lda_o = Dp + 4, lda_h = H + 4;
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* sobs = (float*)smem;                 // [ROWS][Dp+4] kperm
+  float* sx = sobs + ROWS * lda_o;            // [ROWS][H+4] kperm
+  float* shx = sx + ROWS * lda_h;             // [ROWS][H+4] kperm
+  float* shn = shx + ROWS * lda_h;            // [ROWS][H+4] kperm (h')
+  float* shead = shn + ROWS * lda_h;          // [ROWS][HEADC+1]
+
+  for (int ix = tid; ix < ROWS * Dp; ix += 256) {
+    int r = ix / Dp, d = ix % Dp;
+    float vv = 0.0f;
+    if (d < D && row0 + r < p.E) vv = p.obs[(row0 + r) * D + d];
+    sobs[r * lda_o + kperm(d, Dp)] = vv;
+  }
+  for (int ix = tid; ix < ROWS * H; ix += 256) {
+    int r = ix / H, d = ix % H;
+    float vv = 0.0f;
+    int64_t row = row0 + r;
+    if (row < p.E) {
+      vv = p.hx[row * H + d];
+      if (p.done_prev && p.done_prev[row]) vv = 0.0f;
+    }
+    shx[r * lda_h + kperm(d, H)] = vv;
+  }
+  __syncthreads();
+
+  // input layer: x = tanh(obs @ Win + bin)
+  for (int c0 = w * 64; c0 < H; c0 += 256) {
+    f32x4 acc[4] = {};
+    mfma_core_16x64(sobs, lda_o, Dp, p.wtin, H, c0, lane, acc);
+    const int crow = (lane >> 4) * 4, ccol = lane & 15;
+    #pragma unroll
+    for (int n = 0; n < 4; ++n)
+      #pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        int col = c0 + n * 16 + ccol;
+        sx[(crow + i) * lda_h + kperm(col, H)] =
+            tanhf(acc[n][i] + p.bin[col]);
+      }
+  }
+  __syncthreads();
+
+  // gates per 64-col chunk: 6 GEMM accumulations in registers
+  for (int c0 = w * 64; c0 < H; c0 += 256) {
+    f32x4 gir[4] = {}, giz[4] = {}, gin[4] = {};
+    f32x4 ghr[4] = {}, ghz[4] = {}, ghn[4] = {};
+    mfma_core_16x64(sx, lda_h, H, p.wih, 3 * H, 0 * H + c0, lane, gir);
+    mfma_core_16x64(sx, lda_h, H, p.wih, 3 * H, 1 * H + c0, lane, giz);
+    mfma_core_16x64(sx, lda_h, H, p.wih, 3 * H, 2 * H + c0, lane, gin);
+    mfma_core_16x64(shx, lda_h, H, p.whh, 3 * H, 0 * H + c0, lane, ghr);
+    mfma_core_16x64(shx, lda_h, H, p.whh, 3 * H, 1 * H + c0, lane, ghz);
+    mfma_core_16x64(shx, lda_h, H, p.whh, 3 * H, 2 * H + c0, lane, ghn);
+    const int crow = (lane >> 4) * 4, ccol = lane & 15;
+    #pragma unroll
+    for (int n = 0; n < 4; ++n)
+      #pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        int col = c0 + n * 16 + ccol;
+        int r = crow + i;
+        float rr = sigmoidf_(gir[n][i] + p.bih[col]
+                             + ghr[n][i] + p.bhh[col]);
+        float zz = sigmoidf_(giz[n][i] + p.bih[H + col]
+                             + ghz[n][i] + p.bhh[H + col]);
+        float nn = tanhf(gin[n][i] + p.bih[2 * H + col]
+                         + rr * (ghn[n][i] + p.bhh[2 * H + col]));
+        float hold = shx[r * lda_h + kperm(col, H)];
+        float hnew = (1.0f - zz) * nn + zz * hold;
+        shn[r * lda_h + kperm(col, H)] = hnew;
+        int64_t row = row0 + r;
+        if (row < p.E) p.hx_out[row * H + col] = hnew;
+      }
+  }
+  __syncthreads();
+
+  // head + sampling on h' (same as MLP actor)
+  if (w == 0) {
+    f32x4 acc[2] = {};
+    const int kk = lane >> 4, bc = lane & 15;
+    const float* arow = shn + (lane & 15) * lda_h + kk * (H >> 2);
+    const float* bbase = p.wth + (int64_t)4 * (kk * HEADC + bc);
+    for (int k16 = 0; k16 < (H >> 4); ++k16) {
+      f32x4 av = *(const f32x4*)(arow + 4 * k16);
+      const float* bk = bbase + (int64_t)16 * k16 * HEADC;
+      #pragma unroll
+      for (int n = 0; n < 2; ++n) {
+        f32x4 bv = *(const f32x4*)(bk + 64 * n);
+        #pragma unroll
+        for (int s = 0; s < 4; ++s)
+          acc[n] = __builtin_amdgcn_mfma_f32_16x16x4f32(av[s], bv[s], acc[n],
+                                                        0, 0, 0);
+      }
+    }
+    const int crow = (lane >> 4) * 4, ccol = lane & 15;
+    #pragma unroll
+    for (int n = 0; n < 2; ++n)
+      #pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        int col = n * 16 + ccol;
+        shead[(crow + i) * (HEADC + 1) + col] = acc[n][i] + p.bh[col];
+      }
+  }
+  __syncthreads();
+
+  if (w == 0 && lane < ROWS) {
+    int64_t row = row0 + lane;
+    if (row < p.E) {
+      const float* hrow = shead + lane * (HEADC + 1);
+      p.value[row] = hrow[A];
+      if (!p.value_only) {
+        const uint8_t* mrow = p.mask + row * A;
+        float mx = -3.0e38f;
+        for (int i = 0; i < A; ++i)
+          if (mrow[i]) mx = fmaxf(mx, hrow[i]);
+        float sum = 0.0f;
+        for (int i = 0; i < A; ++i)
+          if (mrow[i]) sum += __expf(hrow[i] - mx);
+        float thresh = p.u[row] * sum;
+        float s = 0.0f;
+        int act = -1;
+        for (int i = 0; i < A; ++i) {
+          if (!mrow[i]) continue;
+          s += __expf(hrow[i] - mx);
+          act = i;
+          if (s >= thresh) break;
+        }
+        p.actions[row] = act;
+        p.logp[row] = (hrow[act] - mx) - __logf(sum);
+      }
+    }
+  }
+}
+
+}  // namespace
+
+extern "C" int rlsched_gru_actor(const int64_t* ip, const uint64_t* bufs,
+                                 void* stream) {
+  GruActorArgs p;
+  p.E = ip[0]; p.D = ip[1]; p.H = ip[2]; p.A = ip[3]; p.value_only = ip[4];
+  if (p.H % 64 || p.A + 1 > HEADC) return -1;
+  p.obs = (const float*)bufs[0];
+  p.mask = (const uint8_t*)bufs[1];
+  p.u = (const float*)bufs[2];
+  p.done_prev = (const uint8_t*)bufs[3];   // 0 -> no reset
+  p.hx = (const float*)bufs[4];
+  p.hx_out = (float*)bufs[5];
+  p.wtin = (const float*)bufs[6];
+  p.bin = (const float*)bufs[7];
+  p.wih = (const float*)bufs[8];
+  p.bih = (const float*)bufs[9];
+  p.whh = (const float*)bufs[10];
+  p.bhh = (const float*)bufs[11];
+  p.wth = (const float*)bufs[12];
+  p.bh = (const float*)bufs[13];
+  p.actions = (int64_t*)bufs[14];
+  p.logp = (float*)bufs[15];
+  p.value = (float*)bufs[16];
+  const int Dp = ((int)p.D + 15) & ~15;
+  size_t lds = sizeof(float) * (ROWS * (Dp + 4) + 3 * ROWS * ((int)p.H + 4)
+                                + ROWS * (HEADC + 1));
+  dim3 grid((unsigned)((p.E + ROWS - 1) / ROWS));
+  hipLaunchKernelGGL(gru_actor_kernel, grid, dim3(256), lds,
+                     (hipStream_t)stream, p);
+  return (int)hipGetLastError();
+}

@@ -318,9 +318,11 @@ __global__ __launch_bounds__(256) void bwd_dw_kernel(BwdDwArgs p) {
 }
 
 // ------------------------------------------------------- bias column sums
-// db[N] = sum_b dY[b][n] (mean scaling already inside dY via the loss).
+// db[N] += sum_b dY[b][n] (caller zeroes db). Two-level: each block owns a
+// 256-column tile x one batch chunk (coalesced row reads), one atomicAdd
+// per column per chunk.
 struct ColSumArgs {
-  int64_t B, N;
+  int64_t B, N, BCS;   // batch rows per chunk
   const float* dY;
   float* db;
 };
@@ -328,9 +330,11 @@ struct ColSumArgs {
 __global__ __launch_bounds__(256) void colsum_kernel(ColSumArgs p) {
   int64_t n = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (n >= p.N) return;
+  int64_t b0 = (int64_t)blockIdx.y * p.BCS;
+  int64_t b1 = b0 + p.BCS < p.B ? b0 + p.BCS : p.B;
   float s = 0.f;
-  for (int64_t b = 0; b < p.B; ++b) s += p.dY[b * p.N + n];
-  p.db[n] = s;
+  for (int64_t b = b0; b < b1; ++b) s += p.dY[b * p.N + n];
+  atomicAdd(&p.db[n], s);
 }
 
 }  // namespace
@@ -417,9 +421,11 @@ extern "C" int rlsched_colsum(const int64_t* ip, const uint64_t* bufs,
                               void* stream) {
   ColSumArgs p;
   p.B = ip[0]; p.N = ip[1];
+  p.BCS = 512;
   p.dY = (const float*)bufs[0];
   p.db = (float*)bufs[1];
-  dim3 grid((unsigned)((p.N + 255) / 256));
+  unsigned chunks = (unsigned)((p.B + p.BCS - 1) / p.BCS);
+  dim3 grid((unsigned)((p.N + 255) / 256), chunks);
   hipLaunchKernelGGL(colsum_kernel, grid, dim3(256), 0, (hipStream_t)stream, p);
   return (int)hipGetLastError();
 }

@@ -81,5 +81,6 @@ def bwd_dw(X, dY, dW, bc: int = 1024):
 
 
 def colsum(dY, db):
+    """db += column sums of dY (zero db first)."""
     B, N = dY.shape
     _call("rlsched_colsum", [B, N], _ptrs(dY, db), _stream(dY))

@@ -105,10 +105,19 @@ def train(cfg: TrainConfig, iterations: int | None = None,
     policy = make_policy(cfg, device)
     broadcast_params(policy)
     grad_sync = FlatGradAllreduce(policy) if world > 1 else None
-    collector = Collector(env, policy, cfg.rollout_T, device,
-                          seed=cfg.seed * 101 + rank)
+    on_gpu_fast = device.type == "cuda" and cfg.use_hip_env
+    if on_gpu_fast:
+        from .agents.rollout_gpu import GpuRolloutEngine
+        collector = GpuRolloutEngine(env, policy, cfg.rollout_T,
+                                     seed=cfg.seed * 101 + rank)
+    else:
+        collector = Collector(env, policy, cfg.rollout_T, device,
+                              seed=cfg.seed * 101 + rank)
     if cfg.algo == "random":
         learner = None
+    elif (on_gpu_fast and cfg.algo == "ppo" and not policy.is_recurrent):
+        from .agents.ppo_hip import HipPPOLearner
+        learner = HipPPOLearner(policy, cfg.ppo, grad_sync)
     else:
         learner = make_learner(cfg, policy, grad_sync)
     iters = iterations if iterations is not None else cfg.iterations
