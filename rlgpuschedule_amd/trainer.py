@@ -91,7 +91,8 @@ def make_learner(cfg: TrainConfig, policy, grad_sync=None):
 
 
 def train(cfg: TrainConfig, iterations: int | None = None,
-          log_fn=print) -> dict:
+          log_fn=print, out_dir: str | None = None, save_every: int = 0,
+          resume: str | None = None) -> dict:
     """Synchronous trainer (ppo/a2c). IMPALA lives in agents/impala.py."""
     rank, world, local = init_from_env()
     device = resolve_device(cfg)
@@ -120,15 +121,40 @@ def train(cfg: TrainConfig, iterations: int | None = None,
         learner = HipPPOLearner(policy, cfg.ppo, grad_sync)
     else:
         learner = make_learner(cfg, policy, grad_sync)
+    start_it = 0
+    if resume:
+        from .ckpt.io import load_policy
+        meta = load_policy(resume, policy,
+                           getattr(learner, "opt", None),
+                           map_location=str(device))
+        start_it = int(meta.get("meta", {}).get("iteration", 0))
+        if rank == 0 and log_fn:
+            log_fn(f"[{cfg.name}] resumed from {resume} at iter {start_it}")
+    logger = None
+    if out_dir is not None and rank == 0:
+        from .metrics.logger import MetricsLogger
+        logger = MetricsLogger(out_dir, echo=False,
+                               env_steps_per_iter=cfg.num_envs * cfg.rollout_T)
     iters = iterations if iterations is not None else cfg.iterations
     stats = {}
-    for it in range(iters):
+    from .metrics.logger import check_finite
+    for it in range(start_it, start_it + iters):
         buf = collector.collect()
         if learner is not None:
             stats = learner.update(buf)
-        if rank == 0 and log_fn and (it % 10 == 0 or it == iters - 1):
+            check_finite(stats, where=f"iter {it}")
+        if logger is not None:
+            logger.log(stats, iteration=it)
+        if rank == 0 and log_fn and (it % 10 == 0 or it == start_it + iters - 1):
             log_fn(f"[{cfg.name}] iter {it}: " +
                    " ".join(f"{k}={v:.4f}" for k, v in stats.items()))
+        if (save_every and rank == 0 and out_dir is not None
+                and learner is not None
+                and ((it + 1) % save_every == 0 or it == start_it + iters - 1)):
+            from .ckpt.io import save_policy
+            save_policy(f"{out_dir}/policy_{it + 1:06d}.pt", policy,
+                        spec=cfg.spec, optimizer=getattr(learner, "opt", None),
+                        meta={"iteration": it + 1, "config": cfg.name})
     out = dict(stats)
     if hasattr(env, "episode_stats"):
         out["episode_stats"] = env.episode_stats()
