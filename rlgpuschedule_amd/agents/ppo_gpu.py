@@ -42,7 +42,10 @@ class FusedPPOLearner:
         self.policy = policy
         self.cfg = cfg
         self.grad_sync = grad_sync
-        self.use_graph = use_graph and grad_sync is None
+        # graph capture is attempted even with DP grad sync (RCCL ops are
+        # capturable on recent ROCm); on capture failure we fall back to
+        # eager execution (update() try/except)
+        self.use_graph = use_graph
         # capturable Adam so the optimizer step can live inside the graph
         self.opt = torch.optim.Adam(policy.parameters(), lr=cfg.lr, eps=1e-5,
                                     capturable=self.use_graph)
@@ -108,21 +111,32 @@ class FusedPPOLearner:
                                                cfg.max_grad_norm)
                 self.opt.step()
 
+    def _capture(self, buf: RolloutBuffer):
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            self._update_body(buf)   # warmup (autograd/Adam/comm state)
+        torch.cuda.current_stream().wait_stream(s)
+        torch.cuda.synchronize()
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            self._update_body(buf)
+        self._graph = g
+
     def update(self, buf: RolloutBuffer) -> dict:
         if self._bufs is None:
             self._alloc(buf)
         self._refill_perm()
+        if self.use_graph and self._graph is None:
+            try:
+                self._capture(buf)
+            except Exception as e:
+                import warnings
+                warnings.warn(f"hipGraph capture of the PPO update failed "
+                              f"({e!r}); falling back to eager execution")
+                self.use_graph = False
+                self._graph = None
         if self.use_graph:
-            if self._graph is None:
-                s = torch.cuda.Stream()
-                s.wait_stream(torch.cuda.current_stream())
-                with torch.cuda.stream(s):
-                    self._update_body(buf)   # warmup (autograd + Adam state)
-                torch.cuda.current_stream().wait_stream(s)
-                torch.cuda.synchronize()
-                self._graph = torch.cuda.CUDAGraph()
-                with torch.cuda.graph(self._graph):
-                    self._update_body(buf)
             self._graph.replay()
         else:
             self._update_body(buf)
