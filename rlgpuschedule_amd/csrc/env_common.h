@@ -24,6 +24,7 @@ struct RlsParams {
   int64_t time_scale_us;
   int64_t horizon;
   int64_t gandiva;    // 0/1
+  int64_t elastic;    // 0/1 (ENV_SPEC v3)
 };
 
 // Device buffers; all row-major with env as the leading dim.
@@ -33,6 +34,7 @@ struct RlsBufs {
   const int64_t* trace_arr;      // [J]
   const int32_t* trace_gpus;     // [J]
   const int64_t* trace_work;     // [J]
+  const int32_t* trace_ming;     // [J] min demand (== gpus when rigid)
   const int64_t* job_lo;         // [E]
   const int64_t* job_hi;         // [E]
   const int64_t* base;           // [E]
@@ -65,7 +67,7 @@ struct RlsBufs {
   int64_t* ep_decisions;         // [E]
 };
 
-#define RLS_NUM_BUFS 33
+#define RLS_NUM_BUFS 34
 
 extern "C" {
 // Both return 0 on success, a hipError_t otherwise. `stream` is a

@@ -59,7 +59,7 @@ struct Sc {
 struct EnvView {
   const RlsParams* p;
   const int64_t *trace_arr, *trace_work;
-  const int32_t *trace_gpus;
+  const int32_t *trace_gpus, *trace_ming;
   int64_t job_lo, job_hi, base;
   int32_t *node_free;
   int32_t *pend_job;
@@ -75,6 +75,7 @@ __device__ EnvView make_view(const RlsParams& p, const RlsBufs& b, int64_t e) {
   EnvView v;
   v.p = &p;
   v.trace_arr = b.trace_arr; v.trace_gpus = b.trace_gpus; v.trace_work = b.trace_work;
+  v.trace_ming = b.trace_ming;
   v.job_lo = b.job_lo[e]; v.job_hi = b.job_hi[e]; v.base = b.base[e];
   v.node_free = b.node_free + e * p.N;
   v.pend_job = b.pend_job + e * p.P;
@@ -157,10 +158,36 @@ __device__ void admit(const EnvView& v, Sc& s) {
   }
 }
 
-// Work for pending position pos (override for preempted jobs).
-__device__ __forceinline__ int64_t pend_work(const EnvView& v, int pos) {
+// Remaining work of pending position pos, in GPU-us (ENV_SPEC v3;
+// pend_override stores GPU-us; exact for rigid jobs).
+__device__ __forceinline__ int64_t pend_rem_gpuus(const EnvView& v, int pos) {
   int64_t o = v.pend_override[pos];
-  return o > 0 ? o : v.trace_work[v.pend_job[pos]];
+  if (o > 0) return o;
+  int32_t j = v.pend_job[pos];
+  return v.trace_work[j] * (int64_t)v.trace_gpus[j];
+}
+
+// Obs work feature: remaining wall-time at max demand.
+__device__ __forceinline__ int64_t pend_work_us(const EnvView& v, int pos) {
+  int64_t o = v.pend_override[pos];
+  int32_t j = v.pend_job[pos];
+  if (o > 0) {
+    int64_t g = v.trace_gpus[j];
+    return (o + g - 1) / g;
+  }
+  return v.trace_work[j];
+}
+
+// Descending-g placement scan. Returns allocated g (pairs filled) or -1.
+__device__ int try_fit(const EnvView& v, int32_t j, int* pn, int* pc_,
+                       int* span_out) {
+  int g_max = v.trace_gpus[j];
+  int g_min = v.p->elastic ? v.trace_ming[j] : g_max;
+  for (int g = g_max; g >= g_min; --g) {
+    int span = first_fit(v, g, pn, pc_);
+    if (span >= 0) { *span_out = span; return g; }
+  }
+  return -1;
 }
 
 // Remove pending position a: wave-parallel left shift. Cross-chunk
@@ -189,18 +216,19 @@ __device__ int free_slot(const EnvView& v) {
 // Schedule pending position a. Returns true on success.
 __device__ bool place(const EnvView& v, Sc& s, int a, int lane) {
   int32_t j = v.pend_job[a];
-  int g = v.trace_gpus[j];
   int pn[RLS_MAX_SPAN], pc_[RLS_MAX_SPAN];
-  int span = first_fit(v, g, pn, pc_);
-  if (span < 0) return false;
+  int span;
+  int g = try_fit(v, j, pn, pc_, &span);
+  if (g < 0) return false;
   int slot = free_slot(v);
   if (slot < 0) return false;
   for (int i = 0; i < span; ++i) v.node_free[pn[i]] -= pc_[i];
+  int64_t rem = pend_rem_gpuus(v, a);
   v.run_valid[slot] = 1;
   v.run_job[slot] = j;
-  v.run_gpus[slot] = g;
+  v.run_gpus[slot] = g;   // ALLOCATED gpus (== demand when rigid)
   v.run_start[slot] = s.clock;
-  v.run_finish[slot] = s.clock + pend_work(v, a);
+  v.run_finish[slot] = s.clock + (rem + g - 1) / g;
   const int S = (int)v.p->S;
   for (int i = 0; i < S; ++i)
     v.run_alloc[slot * S + i] = i < span ? ((pn[i] << 16) | pc_[i]) : 0;
@@ -268,8 +296,10 @@ __device__ bool do_migrate(const EnvView& v, Sc& s) {
     if (cnt > 0) v.node_free[old_pk[i] >> 16] += cnt;
   }
   int pn[RLS_MAX_SPAN], pc_[RLS_MAX_SPAN];
-  int span = first_fit(v, v.run_gpus[best_r], pn, pc_);
-  if (span < 0) {  // restore, degrade to NOOP
+  int span;
+  int g_old = v.run_gpus[best_r];
+  int g_new = try_fit(v, v.run_job[best_r], pn, pc_, &span);
+  if (g_new < 0) {  // restore, degrade to NOOP
     for (int i = 0; i < S; ++i) {
       int cnt = old_pk[i] & 0xFFFF;
       if (cnt > 0) v.node_free[old_pk[i] >> 16] -= cnt;
@@ -279,6 +309,13 @@ __device__ bool do_migrate(const EnvView& v, Sc& s) {
   for (int i = 0; i < span; ++i) v.node_free[pn[i]] -= pc_[i];
   for (int i = 0; i < S; ++i)
     v.run_alloc[best_r * S + i] = i < span ? ((pn[i] << 16) | pc_[i]) : 0;
+  if (g_new != g_old) {  // elastic re-placement changes speed
+    int64_t rem0 = s.clock < v.run_finish[best_r]
+                   ? v.run_finish[best_r] - s.clock : 1;
+    int64_t rem = rem0 * g_old;
+    v.run_gpus[best_r] = g_new;
+    v.run_finish[best_r] = s.clock + (rem + g_new - 1) / g_new;
+  }
   return true;
 }
 
@@ -294,6 +331,7 @@ __device__ bool do_timeslice(const EnvView& v, Sc& s) {
   }
   int64_t remaining = v.run_finish[best_r] - s.clock;
   if (remaining < 1) remaining = 1;
+  remaining *= v.run_gpus[best_r];   // GPU-us (ENV_SPEC v3)
   v.pend_job[s.pend_count] = v.run_job[best_r];
   v.pend_override[s.pend_count] = remaining;
   s.pend_count += 1;
@@ -323,12 +361,13 @@ __device__ void write_obs_mask(const EnvView& v, const Sc& s, int lane) {
     if (valid) {
       int32_t j = v.pend_job[lane];
       int g = v.trace_gpus[j];
-      fits = first_fit(v, g, nullptr, nullptr) >= 0;
+      int g_fit = v.p->elastic ? v.trace_ming[j] : g;
+      fits = first_fit(v, g_fit, nullptr, nullptr) >= 0;
       int bix = 4 + 5 * lane;
       v.obs[bix + 0] = 1.0f;
       v.obs[bix + 1] = (float)g / (float)(v.p->N * v.p->G);
       v.obs[bix + 2] = lognorm(s.clock - arrival_of(v, j));
-      v.obs[bix + 3] = lognorm(pend_work(v, lane));
+      v.obs[bix + 3] = lognorm(pend_work_us(v, lane));
       v.obs[bix + 4] = fits ? 1.0f : 0.0f;
     }
     v.mask[lane] = (uint8_t)(valid && fits);
@@ -414,7 +453,7 @@ void unpack(const int64_t* params, const uint64_t* bufs, RlsParams& p, RlsBufs& 
   p.E = params[0]; p.N = params[1]; p.G = params[2]; p.K = params[3];
   p.P = params[4]; p.R = params[5]; p.S = params[6]; p.A = params[7];
   p.obs_dim = params[8]; p.idle_tick_us = params[9]; p.time_scale_us = params[10];
-  p.horizon = params[11]; p.gandiva = params[12];
+  p.horizon = params[11]; p.gandiva = params[12]; p.elastic = params[13];
   uint64_t* dst = reinterpret_cast<uint64_t*>(&b);
   for (int i = 0; i < RLS_NUM_BUFS; ++i) dst[i] = bufs[i];
 }

@@ -41,9 +41,19 @@ class ClusterEnvRef:
     def _arrival(self, j: int) -> int:
         return int(self.trace.arrival_us[j]) - self.base
 
-    def _work(self, j: int, pend_pos: int | None = None) -> int:
+    def _rem_gpuus(self, j: int, pend_pos: int | None = None) -> int:
+        """Remaining work in GPU-us (ENV_SPEC v3; exact for rigid jobs)."""
         if pend_pos is not None and self.pend_override[pend_pos] > 0:
             return int(self.pend_override[pend_pos])
+        return int(self.trace.work_us[j]) * int(self.trace.gpus[j])
+
+    def _pend_work_us(self, pos: int) -> int:
+        """Obs work feature: remaining wall-time at max demand."""
+        j = int(self.pend_job[pos])
+        o = int(self.pend_override[pos])
+        g = int(self.trace.gpus[j])
+        if o > 0:
+            return -(-o // g)   # ceil
         return int(self.trace.work_us[j])
 
     def first_fit(self, g: int) -> list[tuple[int, int]] | None:
@@ -106,20 +116,31 @@ class ClusterEnvRef:
     def _free_total(self) -> int:
         return int(self.node_free.sum())
 
+    def _try_fit(self, j: int) -> tuple[int, list] | None:
+        """Descending-g placement scan (g == min == max for rigid jobs)."""
+        g_max = int(self.trace.gpus[j])
+        g_min = int(self.trace.min_gpus[j]) if self.spec.elastic else g_max
+        for g in range(g_max, g_min - 1, -1):
+            pairs = self.first_fit(g)
+            if pairs is not None:
+                return g, pairs
+        return None
+
     def _place(self, pend_pos: int) -> bool:
         j = int(self.pend_job[pend_pos])
-        g = int(self.trace.gpus[j])
-        pairs = self.first_fit(g)
-        if pairs is None:
+        fit = self._try_fit(j)
+        if fit is None:
             return False
+        g, pairs = fit
         for n, c in pairs:
             self.node_free[n] -= c
         slot = next(r for r in range(self.spec.run_cap) if not self.run_valid[r])
+        rem = self._rem_gpuus(j, pend_pos)
         self.run_valid[slot] = True
         self.run_job[slot] = j
-        self.run_gpus[slot] = g
+        self.run_gpus[slot] = g   # ALLOCATED gpus (== demand for rigid)
         self.run_start[slot] = self.clock
-        self.run_finish[slot] = self.clock + self._work(j, pend_pos)
+        self.run_finish[slot] = self.clock + (-(-rem // g))
         self.run_alloc[slot] = pairs
         self.run_count += 1
         # remove pending position
@@ -166,17 +187,24 @@ class ClusterEnvRef:
         pairs_old = self.run_alloc[slot]
         for n, c in pairs_old:
             self.node_free[n] += c
-        pairs_new = self.first_fit(int(self.run_gpus[slot]))
-        if pairs_new is None:
+        j = int(self.run_job[slot])
+        g_old = int(self.run_gpus[slot])
+        fit = self._try_fit(j)
+        if fit is None:
             # span-constrained first-fit can fail even after freeing (the
             # freed cluster may be more fragmented than at placement time):
             # restore and degrade to NOOP.
             for n, c in pairs_old:
                 self.node_free[n] -= c
             return False
+        g_new, pairs_new = fit
         for n, c in pairs_new:
             self.node_free[n] -= c
         self.run_alloc[slot] = pairs_new
+        if g_new != g_old:  # elastic re-placement changes speed
+            rem = max(1, int(self.run_finish[slot]) - self.clock) * g_old
+            self.run_gpus[slot] = g_new
+            self.run_finish[slot] = self.clock + (-(-rem // g_new))
         return True
 
     def _timeslice(self) -> bool:
@@ -188,7 +216,8 @@ class ClusterEnvRef:
               if self.run_valid[r]]
         _, negr = max(el)
         slot = -negr
-        remaining = max(1, int(self.run_finish[slot]) - self.clock)
+        remaining = (max(1, int(self.run_finish[slot]) - self.clock)
+                     * int(self.run_gpus[slot]))  # GPU-us (ENV_SPEC v3)
         self.pend_job[self.pend_count] = self.run_job[slot]
         self.pend_override[self.pend_count] = remaining
         self.pend_count += 1
@@ -230,7 +259,9 @@ class ClusterEnvRef:
         m = np.zeros(s.num_actions, bool)
         for k in range(min(s.queue_visible, self.pend_count)):
             j = int(self.pend_job[k])
-            m[k] = self.first_fit(int(self.trace.gpus[j])) is not None
+            g_fit = (int(self.trace.min_gpus[j]) if s.elastic
+                     else int(self.trace.gpus[j]))
+            m[k] = self.first_fit(g_fit) is not None
         m[s.noop_action] = True
         if s.gandiva:
             m[s.migrate_action] = self.run_count > 0
@@ -250,12 +281,14 @@ class ClusterEnvRef:
         for k in range(min(s.queue_visible, self.pend_count)):
             j = int(self.pend_job[k])
             g = int(self.trace.gpus[j])
+            g_fit = int(self.trace.min_gpus[j]) if s.elastic else g
             b = 4 + 5 * k
             obs[b + 0] = 1.0
             obs[b + 1] = np.float32(g) / np.float32(C)
             obs[b + 2] = _lognorm(self.clock - self._arrival(j))
-            obs[b + 3] = _lognorm(self._work(j, k))
-            obs[b + 4] = np.float32(1.0 if self.first_fit(g) is not None else 0.0)
+            obs[b + 3] = _lognorm(self._pend_work_us(k))
+            obs[b + 4] = np.float32(
+                1.0 if self.first_fit(g_fit) is not None else 0.0)
         return obs
 
     # integer-state snapshot for differential tests against the HIP env

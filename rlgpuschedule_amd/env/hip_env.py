@@ -44,6 +44,7 @@ class HipVectorEnv:
             "trace_arr": t(pool.arrival_us),
             "trace_gpus": t(pool.gpus),
             "trace_work": t(pool.work_us),
+            "trace_ming": t(pool.min_gpus),
             "job_lo": t(job_lo), "job_hi": t(job_hi),
             "base": t(pool.arrival_us[job_lo]),
             "clock": z(E, torch.int64), "cursor": z(E, torch.int64),
@@ -77,6 +78,7 @@ class HipVectorEnv:
             E, N, spec.gpus_per_node, spec.queue_visible, P, R, S, A, OD,
             spec.idle_tick_us, spec.time_scale_us, spec.horizon,
             1 if spec.gandiva else 0,
+            1 if spec.elastic else 0,
         ], dtype=np.int64)
         self.binding = EnvStepBinding(
             params, {k: v.data_ptr() for k, v in self.b.items()})

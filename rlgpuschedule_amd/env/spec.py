@@ -21,6 +21,7 @@ class EnvSpec:
     time_scale_us: int = 10_000_000
     horizon: int = 50_000
     gandiva: bool = False            # adds MIGRATE/TIMESLICE actions (v2)
+    elastic: bool = False            # elastic multi-GPU jobs (v3)
 
     def __post_init__(self):
         if self.run_cap == 0:

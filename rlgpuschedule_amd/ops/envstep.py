@@ -13,7 +13,8 @@ import numpy as np
 from .build import build_lib
 
 BUF_ORDER = [
-    "trace_arr", "trace_gpus", "trace_work", "job_lo", "job_hi", "base",
+    "trace_arr", "trace_gpus", "trace_work", "trace_ming",
+    "job_lo", "job_hi", "base",
     "clock", "cursor", "node_free", "pend_job", "pend_override", "pend_count",
     "run_valid", "run_job", "run_gpus", "run_finish", "run_start", "run_alloc",
     "run_count", "decisions", "completed", "jct_sum", "reward_sum",
@@ -21,7 +22,7 @@ BUF_ORDER = [
     "ep_count", "ep_return", "ep_jct_sum", "ep_completed", "ep_decisions",
 ]
 NUM_BUFS = len(BUF_ORDER)
-assert NUM_BUFS == 33
+assert NUM_BUFS == 34
 
 _lib = None
 
@@ -43,7 +44,7 @@ def load() -> ctypes.CDLL:
 
 class EnvStepBinding:
     def __init__(self, params: np.ndarray, buf_ptrs: dict[str, int]):
-        assert params.dtype == np.int64 and len(params) == 13
+        assert params.dtype == np.int64 and len(params) == 14
         self.lib = load()
         self.params = np.ascontiguousarray(params)
         self._ptrs = dict(buf_ptrs)

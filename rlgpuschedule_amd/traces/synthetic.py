@@ -24,8 +24,13 @@ _US = 1_000_000
 @dataclass
 class JobTrace:
     arrival_us: np.ndarray  # int64, sorted ascending
-    gpus: np.ndarray        # int32
-    work_us: np.ndarray     # int64 (service time)
+    gpus: np.ndarray        # int32 (max demand)
+    work_us: np.ndarray     # int64 (service time at max demand)
+    min_gpus: np.ndarray | None = None  # int32; None -> == gpus (rigid)
+
+    def __post_init__(self):
+        if self.min_gpus is None:
+            self.min_gpus = self.gpus.copy()
 
     def __len__(self) -> int:
         return len(self.arrival_us)
@@ -53,7 +58,13 @@ def philly_like(
     mean_interarrival_s: float = 60.0,
     median_duration_s: float = 600.0,
     sigma: float = 1.4,
+    fail_frac: float = 0.0,
+    elastic: bool = False,
 ) -> JobTrace:
+    """fail_frac: fraction of jobs that FAIL early (the published Philly
+    trace contains failed jobs): their realized service time is cut to
+    1-20%% of the sampled duration. elastic: multi-GPU jobs accept down to
+    a quarter of their max demand (ENV_SPEC v3)."""
     rng = np.random.default_rng(seed)
     inter = rng.exponential(mean_interarrival_s, size=num_jobs)
     arrival = np.cumsum(inter)
@@ -63,10 +74,17 @@ def philly_like(
     gpus = rng.choice(_PHILLY_SIZES[mask], size=num_jobs, p=probs)
     dur = rng.lognormal(mean=np.log(median_duration_s), sigma=sigma, size=num_jobs)
     dur = np.clip(dur, 30.0, 7 * 24 * 3600.0)
+    if fail_frac > 0:
+        failed = rng.random(num_jobs) < fail_frac
+        cut = rng.uniform(0.01, 0.2, size=num_jobs)
+        dur = np.where(failed, np.maximum(dur * cut, 1.0), dur)
+    ming = np.maximum(1, gpus // 4).astype(np.int32) if elastic \
+        else gpus.astype(np.int32)
     return JobTrace(
         arrival_us=(arrival * _US).astype(np.int64),
         gpus=gpus.astype(np.int32),
         work_us=(dur * _US).astype(np.int64),
+        min_gpus=ming,
     )
 
 
@@ -103,15 +121,17 @@ def make_pool(kind: str, num_envs: int, jobs_per_env: int, seed: int,
     Returns (pool, job_lo[num_envs], job_hi[num_envs]).
     """
     gen = philly_like if kind == "philly" else poisson
-    arrs, gps, wks, lo, hi = [], [], [], [], []
+    arrs, gps, wks, mgs, lo, hi = [], [], [], [], [], []
     off = 0
     for e in range(num_envs):
         t = gen(jobs_per_env, seed * 1_000_003 + e, max_job_gpus, **kw)
         arrs.append(t.arrival_us)
         gps.append(t.gpus)
         wks.append(t.work_us)
+        mgs.append(t.min_gpus)
         lo.append(off)
         off += len(t)
         hi.append(off)
-    pool = JobTrace(np.concatenate(arrs), np.concatenate(gps), np.concatenate(wks))
+    pool = JobTrace(np.concatenate(arrs), np.concatenate(gps),
+                    np.concatenate(wks), np.concatenate(mgs))
     return pool, np.asarray(lo, np.int64), np.asarray(hi, np.int64)

@@ -33,6 +33,7 @@ class TrainConfig:
     ppo: PPOConfig = field(default_factory=PPOConfig)
     a2c: A2CConfig = field(default_factory=A2CConfig)
     use_hip_env: bool = True      # False -> CPU reference env
+    trace_kw: dict = field(default_factory=dict)
 
 
 CONFIGS = {
@@ -50,8 +51,9 @@ CONFIGS = {
         algo="ppo", policy="gru"),
     # Gandiva actions + elastic jobs, A2C
     "c4_gandiva_a2c": TrainConfig(
-        name="c4_gandiva_a2c", spec=cluster64(gandiva=True), num_envs=1024,
-        rollout_T=64, algo="a2c", policy="mlp"),
+        name="c4_gandiva_a2c", spec=cluster64(gandiva=True, elastic=True),
+        num_envs=1024, rollout_T=64, algo="a2c", policy="mlp",
+        trace_kw={"elastic": True, "fail_frac": 0.1}),
     # 1024-GPU Poisson trace, IMPALA-style async actor/learner
     "c5_impala": TrainConfig(
         name="c5_impala", spec=cluster1024(), num_envs=2048, rollout_T=64,
@@ -70,9 +72,11 @@ def make_env(cfg: TrainConfig, device: torch.device, rank: int = 0):
     if cfg.use_hip_env and device.type == "cuda":
         from .env.hip_env import HipVectorEnv
         return HipVectorEnv(cfg.spec, cfg.num_envs, trace_kind=cfg.trace_kind,
-                            jobs_per_env=cfg.jobs_per_env, seed=seed)
+                            jobs_per_env=cfg.jobs_per_env, seed=seed,
+                            **cfg.trace_kw)
     venv = CpuVectorEnv(cfg.spec, cfg.num_envs, trace_kind=cfg.trace_kind,
-                        jobs_per_env=cfg.jobs_per_env, seed=seed)
+                        jobs_per_env=cfg.jobs_per_env, seed=seed,
+                        **cfg.trace_kw)
     return TorchEnvAdapter(venv, device)
 
 

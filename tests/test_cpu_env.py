@@ -211,3 +211,60 @@ def test_queue_overflow_stalls_admission():
     assert env.pend_count == 1 and env.cursor == 2
     env.step(spec.noop_action)  # advance -> complete + admit
     assert env.pend_count == 2
+
+
+def test_elastic_placement_and_speed():
+    """ENV_SPEC v3: an 8-GPU (min 2) job lands on 4 free GPUs and runs 2x
+    as long as at full allocation."""
+    spec = EnvSpec(num_nodes=2, gpus_per_node=4, queue_visible=4,
+                   queue_cap=8, run_cap=8, max_span=2, horizon=100,
+                   elastic=True)
+    tr = JobTrace(
+        arrival_us=np.zeros(2, np.int64),
+        gpus=np.array([4, 8], np.int32),
+        work_us=np.array([10 * US, 10 * US], np.int64),
+        min_gpus=np.array([4, 2], np.int32),
+    )
+    env = ClusterEnvRef(spec, tr)
+    env.step(0)     # 4-gpu rigid job takes node0
+    assert env.node_free.tolist() == [0, 4]
+    mask = env.action_mask()
+    assert mask[0]  # elastic 8-gpu job fits at min 2
+    env.step(0)     # placed at g=4 (largest feasible)
+    assert env.node_free.tolist() == [0, 0]
+    slot = next(r for r in range(spec.run_cap)
+                if env.run_valid[r] and env.run_job[r] == 1)
+    assert env.run_gpus[slot] == 4
+    # rem = 10s * 8 gpus = 80 gpu-s; at g=4 -> 20s wall
+    assert env.run_finish[slot] == 20 * US
+
+
+def test_elastic_timeslice_preserves_gpu_work():
+    spec = EnvSpec(num_nodes=1, gpus_per_node=8, queue_visible=4,
+                   queue_cap=8, run_cap=8, max_span=1, horizon=100,
+                   gandiva=True, elastic=True)
+    tr = JobTrace(
+        arrival_us=np.zeros(1, np.int64),
+        gpus=np.array([8], np.int32),
+        work_us=np.array([10 * US], np.int64),
+        min_gpus=np.array([2], np.int32),
+    )
+    env = ClusterEnvRef(spec, tr)
+    env.step(0)   # runs at g=8, finish=10s
+    assert env.run_finish[0] == 10 * US
+    env.step(spec.timeslice_action)   # preempt at t=0
+    assert env.pend_override[0] == 10 * US * 8  # full gpu-work saved
+    env.step(0)   # re-place (full cluster free -> g=8 again)
+    assert env.run_finish[0] == env.clock + 10 * US
+
+
+def test_rigid_traces_unchanged_by_v3():
+    """pend_override in GPU-us must be invisible for rigid jobs: preempt
+    then re-place reproduces the same finish time."""
+    spec = tiny_spec(gandiva=True)
+    env = ClusterEnvRef(spec, tiny_trace())
+    env.step(0)  # job0: 2 gpus, 10s
+    f0 = int(env.run_finish[0])
+    env.step(spec.timeslice_action)
+    env.step(env.pend_count - 1)  # re-place immediately (clock unchanged)
+    assert int(env.run_finish[0]) == f0

@@ -119,3 +119,13 @@ def test_throughput_smoke():
     rate = 4096 * n / dt
     print(f"env-steps/sec (noop, 4096 envs): {rate:,.0f}")
     assert rate > 10_000
+
+
+@pytest.mark.gpu
+def test_parity_elastic():
+    spec = EnvSpec(num_nodes=8, gpus_per_node=8, queue_visible=12,
+                   queue_cap=64, run_cap=64, max_span=4, horizon=600,
+                   gandiva=True, elastic=True)
+    _run_parity(spec, jobs_per_env=24, steps=500,
+                mean_interarrival_s=20.0, median_duration_s=120.0,
+                elastic=True, fail_frac=0.15)
