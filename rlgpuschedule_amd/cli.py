@@ -25,6 +25,17 @@ def main(argv=None):
     tr.add_argument("--out", default=None, help="metrics/ckpt directory")
     tr.add_argument("--save-every", type=int, default=0)
     tr.add_argument("--resume", default=None, help="checkpoint path")
+    ev = sub.add_parser("eval", help="evaluate a checkpoint or baseline")
+    ev.add_argument("--config", choices=sorted(CONFIGS), default="c2_ppo_mlp")
+    ev.add_argument("--ckpt", default=None, help="policy checkpoint (.pt/.safetensors)")
+    ev.add_argument("--baseline", choices=["fifo", "sjf", "random"],
+                    default=None)
+    ev.add_argument("--envs", type=int, default=32)
+    ev.add_argument("--episodes", type=int, default=32)
+    ev.add_argument("--jobs-per-env", type=int, default=64)
+    ev.add_argument("--seed", type=int, default=1234)
+    ev.add_argument("--sample", action="store_true",
+                    help="sample instead of greedy argmax")
     ls = sub.add_parser("configs", help="list configs")
     args = ap.parse_args(argv)
 
@@ -33,6 +44,30 @@ def main(argv=None):
             print(f"{name}: algo={cfg.algo} policy={cfg.policy} "
                   f"cluster={cfg.spec.num_nodes}x{cfg.spec.gpus_per_node} "
                   f"envs={cfg.num_envs} T={cfg.rollout_T}")
+        return 0
+
+    if args.cmd == "eval":
+        import json
+        from .agents.evaluate import evaluate_baseline, evaluate_policy
+        cfg = CONFIGS[args.config]
+        if args.baseline:
+            st = evaluate_baseline(cfg.spec, args.baseline, args.envs,
+                                   args.episodes,
+                                   trace_kind=cfg.trace_kind,
+                                   jobs_per_env=args.jobs_per_env,
+                                   seed=args.seed, **cfg.trace_kw)
+        else:
+            from .ckpt.io import load_policy
+            from .trainer import make_policy, resolve_device
+            policy = make_policy(cfg, resolve_device(cfg))
+            if args.ckpt:
+                load_policy(args.ckpt, policy)
+            st = evaluate_policy(cfg.spec, policy, args.envs, args.episodes,
+                                 trace_kind=cfg.trace_kind,
+                                 jobs_per_env=args.jobs_per_env,
+                                 seed=args.seed, greedy=not args.sample,
+                                 **cfg.trace_kw)
+        print(json.dumps(st))
         return 0
 
     cfg = CONFIGS[args.config]

@@ -48,3 +48,25 @@ def test_nan_guard_raises():
     with pytest.raises(FloatingPointError):
         check_finite({"pi_loss": float("nan")})
     check_finite({"pi_loss": 1.0})
+
+
+def test_cli_eval_baseline(capsys):
+    rc = cli_main(["eval", "--config", "c1_cpu_random", "--baseline", "sjf",
+                   "--envs", "2", "--episodes", "2", "--jobs-per-env", "12"])
+    assert rc == 0
+    out = json.loads(capsys.readouterr().out.strip().splitlines()[-1])
+    assert out["episodes"] >= 2 and out["mean_jct_us"] > 0
+
+
+def test_cli_eval_policy_cpu(capsys, tmp_path):
+    import dataclasses
+    from rlgpuschedule_amd.ckpt.io import save_policy
+    from rlgpuschedule_amd.trainer import make_policy, resolve_device
+    import torch
+    cfg = CONFIGS["c1_cpu_random"]  # hidden must match what eval builds
+    pol = make_policy(cfg, torch.device("cpu"))
+    save_policy(tmp_path / "p.pt", pol, spec=cfg.spec)
+    rc = cli_main(["eval", "--config", "c1_cpu_random", "--ckpt",
+                   str(tmp_path / "p.pt"), "--envs", "2", "--episodes", "2",
+                   "--jobs-per-env", "10"])
+    assert rc == 0
