@@ -34,6 +34,7 @@ class ImpalaConfig:
     ent_coef: float = 0.01
     max_grad_norm: float = 0.5
     behavior_sync: int = 1   # refresh behavior policy every N updates
+    norm_reward: bool = True
 
 
 class VtraceLearner:
@@ -48,6 +49,11 @@ class VtraceLearner:
 
     def update(self, buf: RolloutBuffer) -> dict:
         cfg = self.cfg
+        if cfg.norm_reward:
+            if not hasattr(self, "_rnorm"):
+                from .rewardnorm import RewardNormalizer
+                self._rnorm = RewardNormalizer(cfg.gamma, buf.reward.device)
+            self._rnorm(buf.reward, buf.done)
         T, E = buf.reward.shape
         B = T * E
         logits, values, _ = self.policy(buf.obs.reshape(B, -1),

@@ -38,6 +38,7 @@ class PPOConfig:
     ent_coef: float = 0.01
     max_grad_norm: float = 0.5
     norm_adv: bool = True
+    norm_reward: bool = True
 
 
 class PPOLearner:
@@ -49,6 +50,11 @@ class PPOLearner:
 
     def update(self, buf: RolloutBuffer) -> dict:
         cfg = self.cfg
+        if cfg.norm_reward:
+            if not hasattr(self, "_rnorm"):
+                from .rewardnorm import RewardNormalizer
+                self._rnorm = RewardNormalizer(cfg.gamma, buf.reward.device)
+            self._rnorm(buf.reward, buf.done)
         adv, ret = compute_gae(buf, cfg.gamma, cfg.lam)
         stats = {"pi_loss": 0.0, "v_loss": 0.0, "entropy": 0.0, "clipfrac": 0.0}
         nup = 0
@@ -156,6 +162,7 @@ class A2CConfig:
     vf_coef: float = 0.5
     ent_coef: float = 0.01
     max_grad_norm: float = 0.5
+    norm_reward: bool = True
 
 
 class A2CLearner:
@@ -170,6 +177,11 @@ class A2CLearner:
 
     def update(self, buf: RolloutBuffer) -> dict:
         cfg = self.cfg
+        if cfg.norm_reward:
+            if not hasattr(self, "_rnorm"):
+                from .rewardnorm import RewardNormalizer
+                self._rnorm = RewardNormalizer(cfg.gamma, buf.reward.device)
+            self._rnorm(buf.reward, buf.done)
         adv, ret = compute_gae(buf, cfg.gamma, cfg.lam)
         T, E = buf.reward.shape
         if self.policy.is_recurrent:

@@ -80,6 +80,11 @@ class FusedPPOLearner:
 
     def _update_body(self, buf: RolloutBuffer):
         cfg = self.cfg
+        if cfg.norm_reward:
+            if not hasattr(self, "_rnorm"):
+                from .rewardnorm import RewardNormalizer
+                self._rnorm = RewardNormalizer(cfg.gamma, buf.reward.device)
+            self._rnorm(buf.reward, buf.done)
         bs = self._bufs
         gae_scan(buf.reward, buf.value, buf.done, buf.last_value,
                  cfg.gamma, cfg.lam, adv=bs["adv"], ret=bs["ret"])

@@ -79,6 +79,11 @@ class HipPPOLearner:
 
     def _update_body(self, buf: RolloutBuffer):
         cfg = self.cfg
+        if cfg.norm_reward:
+            if not hasattr(self, "_rnorm"):
+                from .rewardnorm import RewardNormalizer
+                self._rnorm = RewardNormalizer(cfg.gamma, buf.reward.device)
+            self._rnorm(buf.reward, buf.done)
         s = self._s
         pol = self.policy
         mi = self.mirror

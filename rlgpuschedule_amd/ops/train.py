@@ -73,7 +73,7 @@ def bwd_dx(dY, wtp, h, dX):
     _call("rlsched_bwd_dx", [B, K, N], _ptrs(dY, wtp, h, dX), _stream(dY))
 
 
-def bwd_dw(X, dY, dW, bc: int = 4096):
+def bwd_dw(X, dY, dW, bc: int = 1024):
     """dW[M,N] = X^T @ dY (accumulated; caller zeroes dW first)."""
     B, M = X.shape
     N = dY.shape[1]

@@ -21,7 +21,7 @@ def test_fused_ppo_matches_reference_fullbatch():
     T, E = 16, 64
     pol_a = MLPPolicy(spec.obs_dim, spec.num_actions, 128).cuda()
     pol_b = copy.deepcopy(pol_a)
-    cfg = PPOConfig(epochs=1, minibatches=1)  # no permutation randomness
+    cfg = PPOConfig(epochs=1, minibatches=1, norm_reward=False)
     ref = PPOLearner(pol_a, cfg)
     fused = FusedPPOLearner(pol_b, cfg, use_graph=False)
     buf = _make_buf(T, E, spec, "cuda")

@@ -136,7 +136,7 @@ def test_hip_ppo_learner_matches_reference():
     T, E = 16, 64
     pol_a = MLPPolicy(spec.obs_dim, spec.num_actions, 256).cuda()
     pol_b = copy.deepcopy(pol_a)
-    cfg = PPOConfig(epochs=1, minibatches=1)
+    cfg = PPOConfig(epochs=1, minibatches=1, norm_reward=False)
     ref = PPOLearner(pol_a, cfg)
     hip = HipPPOLearner(pol_b, cfg, use_graph=False)
     buf = _make_buf(T, E, spec, "cuda")
