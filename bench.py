@@ -45,6 +45,7 @@ def main():
     ap.add_argument("--envs", type=int, default=4096, help="envs per GPU")
     ap.add_argument("--rollout", type=int, default=64)
     ap.add_argument("--hidden", type=int, default=256)
+    ap.add_argument("--minibatches", type=int, default=8)
     ap.add_argument("--learner", choices=["hip", "fused"], default="hip",
                     help="hip: fully hand-written update kernels; fused: "
                          "torch GEMMs + fused HIP loss")
@@ -71,7 +72,7 @@ def main():
     pdist.broadcast_params(policy)
     grad_sync = pdist.FlatGradAllreduce(policy) if world > 1 else None
     # Throughput-oriented PPO: 1 epoch x 8 minibatches (documented in config)
-    cfg = PPOConfig(epochs=1, minibatches=8)
+    cfg = PPOConfig(epochs=1, minibatches=args.minibatches)
     if has_gpu:
         from rlgpuschedule_amd.agents.ppo_gpu import FusedPPOLearner
         from rlgpuschedule_amd.agents.ppo_hip import HipPPOLearner
@@ -129,7 +130,8 @@ def main():
                 "global_batch": T * E * n_gpus,
                 "seq_len": T,
                 "envs_per_gpu": E,
-                "ppo": "1 epoch x 8 minibatches, clip 0.2, GAE(0.99, 0.95)",
+                "ppo": f"1 epoch x {args.minibatches} minibatches, "
+                       "clip 0.2, GAE(0.99, 0.95), reward-norm",
                 "update_path": args.learner,
                 "parallelism": f"dp{n_gpus}",
             },
