@@ -42,10 +42,10 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=12)
     ap.add_argument("--warmup", type=int, default=3)
-    ap.add_argument("--envs", type=int, default=4096, help="envs per GPU")
+    ap.add_argument("--envs", type=int, default=8192, help="envs per GPU")
     ap.add_argument("--rollout", type=int, default=64)
     ap.add_argument("--hidden", type=int, default=256)
-    ap.add_argument("--minibatches", type=int, default=8)
+    ap.add_argument("--minibatches", type=int, default=4)
     ap.add_argument("--learner", choices=["hip", "fused"], default="hip",
                     help="hip: fully hand-written update kernels; fused: "
                          "torch GEMMs + fused HIP loss")

@@ -154,3 +154,22 @@ class HipVectorEnv:
                 "decisions": int(h["decisions"][e]),
             })
         return out
+
+    def validate_state(self) -> None:
+        """Invariant checker (a data-race in the wave-lockstep kernel or a
+        bad auto-reset shows up here): GPU-accounting consistency across
+        the whole batch. Host-syncs; call from tests/debug only."""
+        b, s = self.b, self.spec
+        nf = b["node_free"].cpu()
+        assert (nf >= 0).all() and (nf <= s.gpus_per_node).all(), "node_free out of range"
+        rv = (b["run_valid"].cpu() != 0)
+        rc = b["run_count"].cpu()
+        assert (rv.sum(1).to(rc.dtype) == rc).all(), "run_count != #valid slots"
+        pc = b["pend_count"].cpu()
+        assert (pc >= 0).all() and (pc <= s.queue_cap).all(), "pend_count out of range"
+        alloc = (b["run_alloc"].cpu() & 0xFFFF) * rv.unsqueeze(-1)
+        used = alloc.sum((1, 2))
+        free_tot = nf.sum(1)
+        assert (used + free_tot == s.total_gpus).all(), "GPU accounting leak"
+        galloc = b["run_gpus"].cpu() * rv
+        assert (galloc.sum(1) == used).all(), "alloc pairs != allocated gpus"

@@ -129,3 +129,24 @@ def test_parity_elastic():
     _run_parity(spec, jobs_per_env=24, steps=500,
                 mean_interarrival_s=20.0, median_duration_s=120.0,
                 elastic=True, fail_frac=0.15)
+
+
+@pytest.mark.gpu
+def test_state_invariants_under_load():
+    """Race/accounting detector: invariants hold after thousands of
+    randomized batched transitions (incl. auto-resets, gandiva, elastic)."""
+    import torch
+    from rlgpuschedule_amd.env.hip_env import HipVectorEnv
+    spec = EnvSpec(num_nodes=8, gpus_per_node=8, queue_visible=16,
+                   queue_cap=64, run_cap=64, max_span=4, horizon=500,
+                   gandiva=True, elastic=True)
+    env = HipVectorEnv(spec, 512, trace_kind="philly", jobs_per_env=24,
+                       seed=9, elastic=True, fail_frac=0.1)
+    env.reset()
+    g = torch.Generator(device="cuda").manual_seed(0)
+    for i in range(400):
+        acts = torch.randint(0, spec.num_actions, (512,), device="cuda",
+                             generator=g)
+        env.step(acts)
+        if i % 100 == 99:
+            env.validate_state()
