@@ -253,7 +253,8 @@ struct BwdDwArgs {
   float* dW;             // [M,N] accumulated
 };
 
-#define DW_LD 68   // 64 + 4 pad (stride % 64 == 4 -> distinct bank groups)
+#define DW_KC 128      // k-depth per staged chunk
+#define DW_LD (DW_KC + 4)  // row stride: % 64 == 4 -> distinct bank groups
 
 __global__ __launch_bounds__(256) void bwd_dw_kernel(BwdDwArgs p) {
   const int tid = threadIdx.x;
@@ -266,34 +267,43 @@ __global__ __launch_bounds__(256) void bwd_dw_kernel(BwdDwArgs p) {
   const int n0 = (bid % ntiles) * 64; bid /= ntiles;
   const int64_t b0 = (int64_t)bid * p.BC;
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  float* xt = (float*)smem;          // [64 m][DW_LD] kperm over k
+  float* xt = (float*)smem;          // [64 m][DW_LD] kperm over k (128-deep)
   float* yt = xt + 64 * DW_LD;       // [64 n][DW_LD]
   f32x4 acc[4] = {};
   const int wm = w * 16;             // wave's 16 m-rows within the tile
   const int ar = lane & 15, kk = lane >> 4, bc = lane & 15;
-  for (int kc = 0; kc < (int)p.BC; kc += 64) {
-    // stage X^T and dY^T chunks: global row b (=k), LDS row = feature col,
-    // k stored kperm so a lane's 4 k's per macro-step are one float4
-    for (int ix = tid; ix < 64 * 64; ix += 256) {
-      int k = ix >> 6, j = ix & 63;   // k: batch row in chunk, j: feature
+  for (int kc = 0; kc < (int)p.BC; kc += DW_KC) {
+    // stage X^T and dY^T chunks with float4 global loads: thread ix covers
+    // (k, j4): 4 consecutive features j of batch row b -> 4 LDS rows at
+    // column kperm(k). 128 k x 16 j-quads = 2048 quads / 256 threads.
+    for (int ix = tid; ix < DW_KC * 16; ix += 256) {
+      int k = ix >> 4, j4 = (ix & 15) * 4;
       int64_t b = b0 + kc + k;
-      int kp = kperm(k, 64);
-      float xv = 0.f, yv = 0.f;
+      int kp = kperm(k, DW_KC);
+      f32x4 xv = {}, yv = {};
       if (b < p.B) {
-        if (m0 + j < M) xv = p.X[b * M + m0 + j];
-        if (n0 + j < N) yv = p.dY[b * N + n0 + j];
+        if (m0 + j4 + 3 < M) xv = *(const f32x4*)(p.X + b * M + m0 + j4);
+        else
+          for (int t = 0; t < 4; ++t)
+            if (m0 + j4 + t < M) xv[t] = p.X[b * M + m0 + j4 + t];
+        if (n0 + j4 + 3 < N) yv = *(const f32x4*)(p.dY + b * N + n0 + j4);
+        else
+          for (int t = 0; t < 4; ++t)
+            if (n0 + j4 + t < N) yv[t] = p.dY[b * N + n0 + j4 + t];
       }
-      xt[j * DW_LD + kp] = xv;
-      yt[j * DW_LD + kp] = yv;
+      #pragma unroll
+      for (int t = 0; t < 4; ++t) {
+        xt[(j4 + t) * DW_LD + kp] = xv[t];
+        yt[(j4 + t) * DW_LD + kp] = yv[t];
+      }
     }
     __syncthreads();
-    // wave computes C[wm..wm+15][0..63] over this 64-deep k chunk; both
-    // fragments are float4 LDS reads (kperm: k = 16*k16 + 4s + kk ->
-    // position kk*16 + 4*k16 + s)
-    const float* arow = xt + (wm + ar) * DW_LD + kk * 16;
-    const float* brows = yt + bc * DW_LD + kk * 16;
+    // C[wm..wm+15][0..63] over the 128-deep chunk; fragments are float4
+    // LDS reads (kperm: k = 16*k16 + 4s + kk -> kk*(KC/4) + 4*k16 + s)
+    const float* arow = xt + (wm + ar) * DW_LD + kk * (DW_KC / 4);
+    const float* brows = yt + bc * DW_LD + kk * (DW_KC / 4);
     #pragma unroll
-    for (int k16 = 0; k16 < 4; ++k16) {
+    for (int k16 = 0; k16 < DW_KC / 16; ++k16) {
       f32x4 av = *(const f32x4*)(arow + 4 * k16);
       #pragma unroll
       for (int n = 0; n < 4; ++n) {
@@ -411,7 +421,7 @@ extern "C" int rlsched_bwd_dw(const int64_t* ip, const uint64_t* bufs,
                               void* stream) {
   BwdDwArgs p;
   p.B = ip[0]; p.M = ip[1]; p.N = ip[2]; p.BC = ip[3];
-  if (p.BC % 64) return -1;  // batch-row guards handle B % 64 != 0
+  if (p.BC % DW_KC) return -1;  // batch-row guards handle the tail
   p.X = (const float*)bufs[0];
   p.dY = (const float*)bufs[1];
   p.dW = (float*)bufs[2];
