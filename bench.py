@@ -46,6 +46,9 @@ def main():
     ap.add_argument("--rollout", type=int, default=64)
     ap.add_argument("--hidden", type=int, default=256)
     ap.add_argument("--minibatches", type=int, default=4)
+    ap.add_argument("--shards", type=int, default=2,
+                    help="env shards on separate HIP streams (overlaps "
+                         "actor and env kernels across shards)")
     ap.add_argument("--learner", choices=["hip", "fused"], default="hip",
                     help="hip: fully hand-written update kernels; fused: "
                          "torch GEMMs + fused HIP loss")
@@ -60,8 +63,13 @@ def main():
     E, T = args.envs, args.rollout
     if has_gpu:
         from rlgpuschedule_amd.env.hip_env import HipVectorEnv
-        env = HipVectorEnv(spec, E, trace_kind="philly", jobs_per_env=256,
-                           seed=97 + rank)
+        ns = max(1, args.shards)
+        assert E % ns == 0
+        env = [HipVectorEnv(spec, E // ns, trace_kind="philly",
+                            jobs_per_env=256, seed=97 + rank * 31 + s)
+               for s in range(ns)]
+        if ns == 1:
+            env = env[0]
     else:  # CPU fallback so the script runs anywhere (tiny)
         E = min(E, 8)
         env = TorchEnvAdapter(
@@ -107,7 +115,17 @@ def main():
     n_gpus = world if has_gpu else args.gpus
     env_steps = args.steps * T * E * n_gpus  # whole-job aggregate
     value = env_steps / elapsed
-    st = env.episode_stats() if hasattr(env, "episode_stats") else {}
+    if isinstance(env, list):
+        sts = [e.episode_stats() for e in env]
+        comp = sum(s["completed"] for s in sts)
+        st = {
+            "episodes": sum(s["episodes"] for s in sts),
+            "completed": comp,
+            "mean_jct_us": (sum(s["mean_jct_us"] * s["completed"]
+                                for s in sts) / comp) if comp else 0.0,
+        }
+    else:
+        st = env.episode_stats() if hasattr(env, "episode_stats") else {}
     if rank == 0:
         out = {
             "metric": "env_steps_per_sec",
@@ -134,6 +152,7 @@ def main():
                        "clip 0.2, GAE(0.99, 0.95), reward-norm",
                 "update_path": args.learner,
                 "parallelism": f"dp{n_gpus}",
+                "env_shards": args.shards if has_gpu else 1,
             },
         }
         print(json.dumps(out))

@@ -21,7 +21,8 @@ import torch
 
 from ..models.hip_train import HEADC, TrainMirror
 from ..ops.nn import gae_scan
-from ..ops.train import bwd_dw, bwd_dx, colsum, fwd_train, ppo_head
+from ..ops.train import (bwd_dw, bwd_dx, colsum, fwd_train, gather_mb,
+                         ppo_head)
 from .ppo import PPOConfig
 from .rollout import RolloutBuffer
 
@@ -68,9 +69,27 @@ class HipPPOLearner:
             "olp": z(mb), "ov": z(mb), "advm": z(mb), "retm": z(mb),
             "h1": z(mb, H), "h2": z(mb, H), "head": z(mb, HEADC),
             "dhead": z(mb, HEADC), "dh2": z(mb, H), "dh1": z(mb, H),
-            "dW1": z(D, H), "dW2": z(H, H), "dWh": z(H, HEADC),
-            "db1": z(H), "db2": z(H), "dbh": z(HEADC),
         }
+        # grad buffers as views of ONE flat tensor -> a single zero_() per
+        # minibatch instead of six
+        gflat = torch.zeros(D * H + H * H + H * HEADC + 2 * H + HEADC,
+                            device=dev)
+        off = 0
+        def carve(*shape):
+            nonlocal off
+            n = 1
+            for sdim in shape:
+                n *= sdim
+            t = gflat[off:off + n].view(*shape)
+            off += n
+            return t
+        self._s["gflat"] = gflat
+        self._s["dW1"] = carve(D, H)
+        self._s["dW2"] = carve(H, H)
+        self._s["dWh"] = carve(H, HEADC)
+        self._s["db1"] = carve(H)
+        self._s["db2"] = carve(H)
+        self._s["dbh"] = carve(HEADC)
 
     def _refill_perm(self):
         for ep in range(self.cfg.epochs):
@@ -106,13 +125,9 @@ class HipPPOLearner:
         for ep in range(cfg.epochs):
             for i in range(cfg.minibatches):
                 ix = s["perm"][ep, i * self._mb:(i + 1) * self._mb]
-                torch.index_select(obs, 0, ix, out=s["o"])
-                torch.index_select(mask, 0, ix, out=s["m"])
-                torch.index_select(acts, 0, ix, out=s["a"])
-                torch.index_select(old_logp, 0, ix, out=s["olp"])
-                torch.index_select(old_v, 0, ix, out=s["ov"])
-                torch.index_select(adv, 0, ix, out=s["advm"])
-                torch.index_select(ret, 0, ix, out=s["retm"])
+                gather_mb(ix, obs, mask, acts, old_logp, old_v,
+                          adv.contiguous(), ret, s["o"], s["m"], s["a"],
+                          s["olp"], s["ov"], s["advm"], s["retm"])
                 fwd_train(s["o"], mi.wt1, mi.b1, mi.wt2, mi.b2, mi.wth,
                           mi.bh, s["h1"], s["h2"], s["head"])
                 ppo_head(s["head"], s["m"], s["a"], s["olp"], s["ov"],
@@ -120,8 +135,7 @@ class HipPPOLearner:
                          cfg.ent_coef, s["dhead"], s["stats"])
                 bwd_dx(s["dhead"], mi.whT, s["h2"], s["dh2"])
                 bwd_dx(s["dh2"], mi.w2T, s["h1"], s["dh1"])
-                for k in ("dW1", "dW2", "dWh", "db1", "db2", "dbh"):
-                    s[k].zero_()
+                s["gflat"].zero_()
                 bwd_dw(s["h2"], s["dhead"], s["dWh"])
                 bwd_dw(s["h1"], s["dh2"], s["dW2"])
                 bwd_dw(s["o"], s["dh1"], s["dW1"])

@@ -20,7 +20,14 @@ from .rollout import RolloutBuffer
 class GpuRolloutEngine:
     def __init__(self, env, policy, T: int, seed: int = 0,
                  use_graph: bool = True, num_bufs: int = 1):
-        self.env = env
+        """`env` may be a LIST of HipVectorEnvs (shards): each shard rolls
+        out on its own HIP stream so shard A's env_step overlaps shard B's
+        actor kernel (the per-step chain is serial within a shard). The
+        rollout buffer covers all shards (shard s owns env-rows
+        [off_s, off_{s+1}) — contiguous slices, zero-copy for the learner).
+        """
+        self.envs = env if isinstance(env, (list, tuple)) else [env]
+        self.env = self.envs[0]
         self.policy = policy
         self.T = T
         self.device = torch.device("cuda")
@@ -30,8 +37,16 @@ class GpuRolloutEngine:
         acls = HipActorGRU if policy.is_recurrent else HipActorMLP
         self.actors = [acls(policy) for _ in range(num_bufs)]
         self.actor = self.actors[0]
-        E = env.num_envs
-        spec = env.spec
+        E = sum(e.num_envs for e in self.envs)
+        self.offsets = []
+        off = 0
+        for e in self.envs:
+            self.offsets.append((off, off + e.num_envs))
+            off += e.num_envs
+        self._streams = [torch.cuda.Stream() for _ in self.envs] \
+            if len(self.envs) > 1 else None
+        self._join = torch.cuda.Event() if len(self.envs) > 1 else None
+        spec = self.env.spec
         hid = policy.hidden if policy.is_recurrent else 0
         self.bufs = [RolloutBuffer(T, E, spec.obs_dim, spec.num_actions,
                                    self.device, hidden=hid)
@@ -52,40 +67,51 @@ class GpuRolloutEngine:
         self._graphs: dict[int, torch.cuda.CUDAGraph] = {}
         # initial reset: obs_last/mask_last always hold the NEXT rollout's
         # start state
-        env.reset()
-        self.obs_last.copy_(env.b["obs"])
-        self.mask_last.copy_(env.b["mask"])
+        for sh, (o0, o1) in zip(self.envs, self.offsets):
+            sh.reset()
+            self.obs_last[o0:o1].copy_(sh.b["obs"])
+            self.mask_last[o0:o1].copy_(sh.b["mask"])
 
-    def _rollout_body(self, buf: RolloutBuffer, actor):
-        env, T = self.env, self.T
+    def _rollout_body(self, buf: RolloutBuffer, actor, shard: int = 0):
+        env = self.envs[shard]
+        o0, o1 = self.offsets[shard]
+        T = self.T
         recurrent = self.policy.is_recurrent
         if recurrent:
             # learner replays from h0 (already reset where done_last)
-            buf.h0.copy_(self.hx
-                         * (1.0 - self.done_last.float()).unsqueeze(-1))
+            buf.h0[o0:o1].copy_(
+                self.hx[o0:o1]
+                * (1.0 - self.done_last[o0:o1].float()).unsqueeze(-1))
         for t in range(T):
             if recurrent:
-                dprev = self.done_last if t == 0 else buf.done[t - 1]
-                actor.act(buf.obs[t], buf.mask[t], self.u_all[t], dprev,
-                          self.hx, self.hx, buf.actions[t], buf.logp[t],
-                          buf.value[t])
+                dprev = (self.done_last[o0:o1] if t == 0
+                         else buf.done[t - 1, o0:o1])
+                actor.act(buf.obs[t, o0:o1], buf.mask[t, o0:o1],
+                          self.u_all[t, o0:o1], dprev,
+                          self.hx[o0:o1], self.hx[o0:o1],
+                          buf.actions[t, o0:o1], buf.logp[t, o0:o1],
+                          buf.value[t, o0:o1])
             else:
-                actor.act(buf.obs[t], buf.mask[t], self.u_all[t],
-                          buf.actions[t], buf.logp[t], buf.value[t])
+                actor.act(buf.obs[t, o0:o1], buf.mask[t, o0:o1],
+                          self.u_all[t, o0:o1], buf.actions[t, o0:o1],
+                          buf.logp[t, o0:o1], buf.value[t, o0:o1])
             last = t == T - 1
             env.step(
-                buf.actions[t],
-                obs_out=self.obs_last if last else buf.obs[t + 1],
-                mask_out=self.mask_last if last else buf.mask[t + 1],
-                rew_out=buf.reward[t],
-                done_out=buf.done[t],
+                buf.actions[t, o0:o1],
+                obs_out=self.obs_last[o0:o1] if last else buf.obs[t + 1, o0:o1],
+                mask_out=(self.mask_last[o0:o1] if last
+                          else buf.mask[t + 1, o0:o1]),
+                rew_out=buf.reward[t, o0:o1],
+                done_out=buf.done[t, o0:o1],
             )
         if recurrent:
-            self.done_last.copy_(buf.done[T - 1])
-            actor.values(self.obs_last, self.mask_last, self.done_last,
-                         self.hx, self.hx_scratch, buf.last_value)
+            self.done_last[o0:o1].copy_(buf.done[T - 1, o0:o1])
+            actor.values(self.obs_last[o0:o1], self.mask_last[o0:o1],
+                         self.done_last[o0:o1], self.hx[o0:o1],
+                         self.hx_scratch[o0:o1], buf.last_value[o0:o1])
         else:
-            actor.values(self.obs_last, buf.last_value, self.mask_last)
+            actor.values(self.obs_last[o0:o1], buf.last_value[o0:o1],
+                         self.mask_last[o0:o1])
 
     def collect(self, idx: int = 0, refresh: bool = True) -> RolloutBuffer:
         """Run one T-step rollout into bufs[idx] on the CURRENT stream.
@@ -100,23 +126,38 @@ class GpuRolloutEngine:
         self.u_all.uniform_(generator=self.gen)
         buf.obs[0].copy_(self.obs_last)
         buf.mask[0].copy_(self.mask_last)
+        nsh = len(self.envs)
         if self.use_graph:
-            if idx not in self._graphs:
-                s = torch.cuda.Stream()
-                s.wait_stream(torch.cuda.current_stream())
-                with torch.cuda.stream(s):
-                    self._rollout_body(buf, actor)   # warmup
-                torch.cuda.current_stream().wait_stream(s)
-                # warmup advanced the env: re-chain the replay's start state
-                buf.obs[0].copy_(self.obs_last)
-                buf.mask[0].copy_(self.mask_last)
-                torch.cuda.synchronize()
-                g = torch.cuda.CUDAGraph()
-                with torch.cuda.graph(g):
-                    self._rollout_body(buf, actor)
-                self._graphs[idx] = g
-            self._graphs[idx].replay()
+            for shard in range(nsh):
+                key = (idx, shard)
+                if key not in self._graphs:
+                    s = torch.cuda.Stream()
+                    s.wait_stream(torch.cuda.current_stream())
+                    with torch.cuda.stream(s):
+                        self._rollout_body(buf, actor, shard)   # warmup
+                    torch.cuda.current_stream().wait_stream(s)
+                    # warmup advanced the env: re-chain the start state
+                    o0, o1 = self.offsets[shard]
+                    buf.obs[0, o0:o1].copy_(self.obs_last[o0:o1])
+                    buf.mask[0, o0:o1].copy_(self.mask_last[o0:o1])
+                    torch.cuda.synchronize()
+                    g = torch.cuda.CUDAGraph()
+                    with torch.cuda.graph(g):
+                        self._rollout_body(buf, actor, shard)
+                    self._graphs[key] = g
+            if nsh == 1:
+                self._graphs[(idx, 0)].replay()
+            else:
+                cur = torch.cuda.current_stream()
+                for shard in range(nsh):
+                    ss = self._streams[shard]
+                    ss.wait_stream(cur)
+                    with torch.cuda.stream(ss):
+                        self._graphs[(idx, shard)].replay()
+                for ss in self._streams:
+                    cur.wait_stream(ss)
         else:
-            self._rollout_body(buf, actor)
+            for shard in range(nsh):
+                self._rollout_body(buf, actor, shard)
         self.buf = buf
         return buf

@@ -84,3 +84,20 @@ def colsum(dY, db):
     """db += column sums of dY (zero db first)."""
     B, N = dY.shape
     _call("rlsched_colsum", [B, N], _ptrs(dY, db), _stream(dY))
+
+
+def gather_mb(ix, obs, mask, acts, olp, ov, adv, ret,
+              o_obs, o_mask, o_acts, o_olp, o_ov, o_adv, o_ret):
+    """Gather a PPO minibatch (7 arrays) in ONE kernel launch."""
+    lib = load()
+    if not hasattr(lib, "_gather_ready"):
+        lib.rlsched_gather_mb.restype = ctypes.c_int
+        lib.rlsched_gather_mb.argtypes = [ctypes.c_void_p] * 3
+        lib._gather_ready = True
+    MB = ix.shape[0]
+    D = obs.shape[1]
+    A = mask.shape[1]
+    _call("rlsched_gather_mb", [MB, D, A],
+          _ptrs(ix, obs, mask, acts, olp, ov, adv, ret,
+                o_obs, o_mask, o_acts, o_olp, o_ov, o_adv, o_ret),
+          _stream(obs))

@@ -112,3 +112,31 @@ def test_gpu_rollout_engine_and_graph():
         assert (chosen == 1).all()
     assert torch.isfinite(buf.logp).all()
     assert (buf.reward <= 0).all()  # env rewards are non-positive
+
+
+@pytest.mark.gpu
+def test_sharded_rollout_engine():
+    """2-shard engine (2 HIP streams): legality + finiteness, graph mode."""
+    import torch
+    from rlgpuschedule_amd.agents.ppo import PPOConfig
+    from rlgpuschedule_amd.agents.ppo_hip import HipPPOLearner
+    from rlgpuschedule_amd.agents.rollout_gpu import GpuRolloutEngine
+    from rlgpuschedule_amd.env.hip_env import HipVectorEnv
+    from rlgpuschedule_amd.models.policy import MLPPolicy
+    spec = cluster64(horizon=100_000)
+    envs = [HipVectorEnv(spec, 128, trace_kind="philly", jobs_per_env=32,
+                         seed=11 + s) for s in range(2)]
+    policy = MLPPolicy(spec.obs_dim, spec.num_actions, 256).cuda()
+    eng = GpuRolloutEngine(envs, policy, T=16, seed=0, use_graph=True)
+    learner = HipPPOLearner(policy, PPOConfig(epochs=1, minibatches=2),
+                            use_graph=False)
+    for _ in range(3):
+        buf = eng.collect()
+        stats = learner.update(buf)
+    assert np.isfinite(stats["pi_loss"])
+    for t in range(16):
+        chosen = buf.mask[t].gather(1, buf.actions[t].unsqueeze(1))
+        assert (chosen == 1).all()
+    assert torch.isfinite(buf.logp).all()
+    for e in envs:
+        e.validate_state()
