@@ -1,0 +1,67 @@
+"""RL-quality experiment: train PPO on the 64-GPU sim, then evaluate mean
+JCT against FIFO/SJF/random on IDENTICAL trace pools (same seeds).
+
+Usage (GPU box): python bench/train_eval_jct.py [--iters 300]
+Writes results JSON to stdout; used for docs/PERF.md quality table.
+"""
+import argparse
+import dataclasses
+import json
+import sys
+import time
+from pathlib import Path
+
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
+
+import torch  # noqa: E402
+
+from rlgpuschedule_amd.agents.evaluate import (evaluate_baseline,  # noqa: E402
+                                               evaluate_policy)
+from rlgpuschedule_amd.env.spec import cluster64  # noqa: E402
+from rlgpuschedule_amd.trainer import CONFIGS, make_policy  # noqa: E402
+
+
+def run(iters=300, envs=2048, eval_envs=16, eval_eps=16, jobs=96,
+        interarrival=90.0, duration=600.0, seed=7):
+    """Programmatic version returning the full comparison dict."""
+    from rlgpuschedule_amd.agents.ppo import PPOConfig
+    from rlgpuschedule_amd.agents.ppo_hip import HipPPOLearner
+    from rlgpuschedule_amd.agents.rollout_gpu import GpuRolloutEngine
+    from rlgpuschedule_amd.env.hip_env import HipVectorEnv
+
+    tkw = dict(mean_interarrival_s=interarrival, median_duration_s=duration)
+    spec = cluster64(horizon=200_000)
+    torch.manual_seed(seed)
+    env = HipVectorEnv(spec, envs, trace_kind="philly", jobs_per_env=jobs,
+                       seed=seed, **tkw)
+    policy = make_policy(CONFIGS["c2_ppo_mlp"], torch.device("cuda"))
+    learner = HipPPOLearner(policy, PPOConfig(epochs=1, minibatches=4))
+    engine = GpuRolloutEngine(env, policy, 64, seed=seed)
+    t0 = time.perf_counter()
+    stats = {}
+    for it in range(iters):
+        buf = engine.collect()
+        stats = learner.update(buf)
+    torch.cuda.synchronize()
+    res = {"train_s": round(time.perf_counter() - t0, 1),
+           "final": {k: round(v, 4) for k, v in stats.items()}}
+    ekw = dict(num_envs=eval_envs, min_episodes=eval_eps,
+               jobs_per_env=jobs, seed=991, **tkw)
+    res["ppo_greedy"] = evaluate_policy(spec, policy, greedy=True,
+                                        device="cuda", **ekw)
+    res["ppo_sample"] = evaluate_policy(spec, policy, greedy=False,
+                                        device="cuda", **ekw)
+    for b in ("sjf", "fifo", "random"):
+        res[b] = evaluate_baseline(spec, b, **ekw)
+    for k in ("ppo_greedy", "ppo_sample", "sjf", "fifo", "random"):
+        res[k] = {kk: (round(vv, 1) if isinstance(vv, float) else vv)
+                  for kk, vv in res[k].items()}
+    return res
+
+
+if __name__ == "__main__":
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--iters", type=int, default=300)
+    ap.add_argument("--envs", type=int, default=2048)
+    a = ap.parse_args()
+    print(json.dumps(run(iters=a.iters, envs=a.envs), indent=1))
