@@ -1,0 +1,371 @@
+// Hand-written CDNA4 GRU training kernels (BPTT) for the recurrent PPO
+// update (BASELINE config 3). The GRU recurrence is ROW-LOCAL, so a
+// workgroup owns 16 minibatch envs END-TO-END across all T time steps:
+// the whole sequence forward is ONE kernel launch (gru_seq_fwd) and the
+// whole reverse-mode sweep another (gru_seq_bwd) — no grid syncs, no
+// per-step launches (the torch GRUCell replay costs hundreds of launches
+// per minibatch). Weight gradients contract over batch*time AFTER the
+// sweeps using the generic batched GEMM kernels (bwd_dw/bwd_dx/colsum in
+// train_kernels.hip).
+//
+// Cell math matches torch.nn.GRUCell exactly (gate order r,z,n):
+//   x   = tanh(obs @ Win + bin)                      (input layer)
+//   r   = sigmoid(gi_r + gh_r)   gi = x @ Wih + bih   (gh likewise)
+//   z   = sigmoid(gi_z + gh_z)
+//   n   = tanh(gi_n + r * gh_n)
+//   h'  = (1-z)*n + z*h
+// The step-t hidden input is h~ = h_{t-1} * (1 - done[t-1]) (episode
+// boundary reset), mirroring the rollout engine and the torch replay.
+//
+// Saved forward state (plain [T,Emb,H] layouts for the batched GEMMs):
+//   xs, rs, zs, ns, ghns (= gh_n incl. bias), hprevs (= h~ input), hs.
+// Backward emits dgi_all/dgh_all [T,Emb,3H] and consumes dh_ext (the
+// head gradient w.r.t. h_t).
+//
+// Gradient-check tests: tests/test_gru_train.py vs torch autograd.
+
+#include "nn_common.h"
+
+#define ROWS 16
+
+namespace {
+
+__device__ __forceinline__ float sigf(float x) {
+  return 1.0f / (1.0f + __expf(-x));
+}
+
+struct GruFwdArgs {
+  int64_t T, Emb, D, H;
+  const float* obs;      // [T,Emb,D]
+  const uint8_t* done;   // [T,Emb]
+  const float* h0;       // [Emb,H]
+  const float* wtin;     // packed [Dp,H]
+  const float* bin;      // [H]
+  const float* wih;      // packed [H,3H]
+  const float* bih;      // [3H]
+  const float* whh;      // packed [H,3H]
+  const float* bhh;      // [3H]
+  float* xs;             // [T,Emb,H] out
+  float* rs;             // [T,Emb,H]
+  float* zs;             // [T,Emb,H]
+  float* ns;             // [T,Emb,H]
+  float* ghns;           // [T,Emb,H]
+  float* hprevs;         // [T,Emb,H]
+  float* hs;             // [T,Emb,H]
+};
+
+__global__ __launch_bounds__(256) void gru_seq_fwd_kernel(GruFwdArgs p) {
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int w = tid >> 6;
+  const int64_t r0 = (int64_t)blockIdx.x * ROWS;
+  const int D = (int)p.D, H = (int)p.H;
+  const int Dp = (D + 15) & ~15;
+  const int lda_o = Dp + 4, lda_h = H + 4;
+  const int64_t Emb = p.Emb;
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* sobs = (float*)smem;               // [ROWS][Dp+4] kperm
+  float* sx = sobs + ROWS * lda_o;          // [ROWS][H+4] kperm
+  float* shx = sx + ROWS * lda_h;           // [ROWS][H+4] kperm (h~ input)
+  float* snew = shx + ROWS * lda_h;         // [ROWS][H+4] kperm (h~ next)
+
+  // h~_0 = h0 (pre-masked by the collector); also record hprevs[0]
+  for (int ix = tid; ix < ROWS * H; ix += 256) {
+    int r = ix / H, d = ix % H;
+    float vv = 0.0f;
+    if (r0 + r < Emb) {
+      vv = p.h0[(r0 + r) * H + d];
+      p.hprevs[(r0 + r) * H + d] = vv;
+    }
+    shx[r * lda_h + kperm(d, H)] = vv;
+  }
+  __syncthreads();
+
+  for (int64_t t = 0; t < p.T; ++t) {
+    const int64_t base = t * Emb + r0;          // row offset in [T*Emb]
+    // stage obs[t] tile
+    for (int ix = tid; ix < ROWS * Dp; ix += 256) {
+      int r = ix / Dp, d = ix % Dp;
+      float vv = 0.0f;
+      if (d < D && r0 + r < Emb) vv = p.obs[(base + r) * D + d];
+      sobs[r * lda_o + kperm(d, Dp)] = vv;
+    }
+    __syncthreads();
+    // input layer: x = tanh(obs @ Win + bin); save xs plain
+    for (int c0 = w * 64; c0 < H; c0 += 256) {
+      f32x4 acc[4] = {};
+      mfma_core_16x64(sobs, lda_o, Dp, p.wtin, H, c0, lane, acc);
+      const int crow = (lane >> 4) * 4, ccol = lane & 15;
+      #pragma unroll
+      for (int n = 0; n < 4; ++n)
+        #pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          int col = c0 + n * 16 + ccol;
+          float vv = tanhf(acc[n][i] + p.bin[col]);
+          sx[(crow + i) * lda_h + kperm(col, H)] = vv;
+          if (r0 + crow + i < Emb)
+            p.xs[(base + crow + i) * H + col] = vv;
+        }
+    }
+    __syncthreads();
+    // gates per 64-col chunk; save r,z,n,ghn,h plain; update shx for t+1
+    for (int c0 = w * 64; c0 < H; c0 += 256) {
+      f32x4 gir[4] = {}, giz[4] = {}, gin[4] = {};
+      f32x4 ghr[4] = {}, ghz[4] = {}, ghn[4] = {};
+      mfma_core_16x64(sx, lda_h, H, p.wih, 3 * H, 0 * H + c0, lane, gir);
+      mfma_core_16x64(sx, lda_h, H, p.wih, 3 * H, 1 * H + c0, lane, giz);
+      mfma_core_16x64(sx, lda_h, H, p.wih, 3 * H, 2 * H + c0, lane, gin);
+      mfma_core_16x64(shx, lda_h, H, p.whh, 3 * H, 0 * H + c0, lane, ghr);
+      mfma_core_16x64(shx, lda_h, H, p.whh, 3 * H, 1 * H + c0, lane, ghz);
+      mfma_core_16x64(shx, lda_h, H, p.whh, 3 * H, 2 * H + c0, lane, ghn);
+      const int crow = (lane >> 4) * 4, ccol = lane & 15;
+      #pragma unroll
+      for (int n = 0; n < 4; ++n)
+        #pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          int col = c0 + n * 16 + ccol;
+          int r = crow + i;
+          int64_t grow = base + r;
+          float rr = sigf(gir[n][i] + p.bih[col]
+                          + ghr[n][i] + p.bhh[col]);
+          float zz = sigf(giz[n][i] + p.bih[H + col]
+                          + ghz[n][i] + p.bhh[H + col]);
+          float gn = ghn[n][i] + p.bhh[2 * H + col];
+          float nn = tanhf(gin[n][i] + p.bih[2 * H + col] + rr * gn);
+          float hold = shx[r * lda_h + kperm(col, H)];
+          float hnew = (1.0f - zz) * nn + zz * hold;
+          if (r0 + r < Emb) {
+            p.rs[grow * H + col] = rr;
+            p.zs[grow * H + col] = zz;
+            p.ns[grow * H + col] = nn;
+            p.ghns[grow * H + col] = gn;
+            p.hs[grow * H + col] = hnew;
+          }
+          // h~ for step t+1 (episode reset via done[t]). Other waves may
+          // still be READING sx/shx in their gate K-loops, so the new
+          // state goes to the dedicated snew buffer; shx is swapped in
+          // after the barrier.
+          float hmask = hnew;
+          if (r0 + r < Emb && p.done[t * Emb + r0 + r]) hmask = 0.0f;
+          snew[r * lda_h + kperm(col, H)] = hmask;
+          if (t + 1 < p.T && r0 + r < Emb)
+            p.hprevs[((t + 1) * Emb + r0 + r) * H + col] = hmask;
+        }
+    }
+    __syncthreads();
+    for (int ix = tid; ix < ROWS * lda_h; ix += 256)
+      shx[ix] = snew[ix];
+    __syncthreads();
+  }
+}
+
+struct GruBwdArgs {
+  int64_t T, Emb, H;
+  const float* dh_ext;   // [T,Emb,H] head grad wrt h_t
+  const uint8_t* done;   // [T,Emb]
+  const float* rs;
+  const float* zs;
+  const float* ns;
+  const float* ghns;
+  const float* hprevs;
+  const float* whh_bwd;  // packed [3H,H] (= torch weight_hh as-is)
+  float* dgi_all;        // [T,Emb,3H] out
+  float* dgh_all;        // [T,Emb,3H] out
+};
+
+__global__ __launch_bounds__(256) void gru_seq_bwd_kernel(GruBwdArgs p) {
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int w = tid >> 6;
+  const int64_t r0 = (int64_t)blockIdx.x * ROWS;
+  const int H = (int)p.H;
+  const int H3 = 3 * H;
+  const int lda_h = H + 4, lda_3 = H3 + 4;
+  const int64_t Emb = p.Emb;
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* sdgh = (float*)smem;                 // [ROWS][3H+4] kperm
+  float* scarry = sdgh + ROWS * lda_3;        // [ROWS][H+4] plain: dh carry
+  float* stmp = scarry + ROWS * lda_h;        // [ROWS][H+4] plain: dh*z
+
+  for (int ix = tid; ix < ROWS * lda_h; ix += 256) {
+    scarry[ix] = 0.0f;
+    stmp[ix] = 0.0f;
+  }
+  __syncthreads();
+
+  for (int64_t t = p.T - 1; t >= 0; --t) {
+    const int64_t base = t * Emb + r0;
+    // phase 1: elementwise gate grads; fills sdgh (kperm) + stmp (dh*z)
+    for (int ix = tid; ix < ROWS * H; ix += 256) {
+      int r = ix / H, col = ix % H;
+      int64_t grow = base + r;
+      float dh = scarry[r * lda_h + col];
+      float dgr = 0.f, dgz = 0.f, dgn = 0.f, dghn = 0.f, dhz = 0.f;
+      if (r0 + r < Emb) {
+        dh += p.dh_ext[grow * H + col];
+        float rr = p.rs[grow * H + col];
+        float zz = p.zs[grow * H + col];
+        float nn = p.ns[grow * H + col];
+        float gn = p.ghns[grow * H + col];
+        float hp = p.hprevs[grow * H + col];
+        float dz = dh * (hp - nn);
+        float dn = dh * (1.0f - zz);
+        dhz = dh * zz;
+        float dn_pre = dn * (1.0f - nn * nn);
+        float dr = dn_pre * gn;
+        float dr_pre = dr * rr * (1.0f - rr);
+        float dz_pre = dz * zz * (1.0f - zz);
+        dgr = dr_pre;
+        dgz = dz_pre;
+        dgn = dn_pre;
+        dghn = dn_pre * rr;
+        p.dgi_all[grow * H3 + col] = dgr;
+        p.dgi_all[grow * H3 + H + col] = dgz;
+        p.dgi_all[grow * H3 + 2 * H + col] = dgn;
+        p.dgh_all[grow * H3 + col] = dgr;
+        p.dgh_all[grow * H3 + H + col] = dgz;
+        p.dgh_all[grow * H3 + 2 * H + col] = dghn;
+      }
+      stmp[r * lda_h + col] = dhz;
+      sdgh[r * lda_3 + kperm(col, H3)] = dgr;
+      sdgh[r * lda_3 + kperm(H + col, H3)] = dgz;
+      sdgh[r * lda_3 + kperm(2 * H + col, H3)] = dghn;
+    }
+    __syncthreads();
+    // phase 2: carry GEMM dh_prev = sdgh @ whh_bwd; combine with dh*z and
+    // the episode-boundary mask (done[t-1])
+    for (int c0 = w * 64; c0 < H; c0 += 256) {
+      f32x4 acc[4] = {};
+      mfma_core_16x64(sdgh, lda_3, H3, p.whh_bwd, H, c0, lane, acc);
+      const int crow = (lane >> 4) * 4, ccol = lane & 15;
+      #pragma unroll
+      for (int n = 0; n < 4; ++n)
+        #pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          int col = c0 + n * 16 + ccol;
+          int r = crow + i;
+          float dprev = acc[n][i] + stmp[r * lda_h + col];
+          if (t > 0 && r0 + r < Emb
+              && p.done[(t - 1) * Emb + r0 + r]) dprev = 0.0f;
+          scarry[r * lda_h + col] = dprev;
+        }
+    }
+    __syncthreads();
+  }
+}
+
+// --------------------------------------- standalone head forward (batched)
+// head[B,HEADC] = h[B,H] @ wth + bh (h staged into kperm LDS per 16 rows).
+struct HeadFwdArgs {
+  int64_t B, H;
+  const float* h;       // [B,H]
+  const float* wth;     // packed [H,32]
+  const float* bh;      // [32]
+  float* head;          // [B,32]
+};
+
+#define HEADC 32
+
+__global__ __launch_bounds__(256) void head_fwd_kernel(HeadFwdArgs p) {
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int w = tid >> 6;
+  const int64_t r0 = (int64_t)blockIdx.x * ROWS;
+  const int H = (int)p.H;
+  const int lda_h = H + 4;
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* sh = (float*)smem;   // [ROWS][H+4] kperm
+  for (int ix = tid; ix < ROWS * H; ix += 256) {
+    int r = ix / H, d = ix % H;
+    float vv = 0.0f;
+    if (r0 + r < p.B) vv = p.h[(r0 + r) * H + d];
+    sh[r * lda_h + kperm(d, H)] = vv;
+  }
+  __syncthreads();
+  if (w == 0) {
+    f32x4 acc[2] = {};
+    const int kk = lane >> 4, bc = lane & 15;
+    const float* arow = sh + (lane & 15) * lda_h + kk * (H >> 2);
+    const float* bbase = p.wth + (int64_t)4 * (kk * HEADC + bc);
+    for (int k16 = 0; k16 < (H >> 4); ++k16) {
+      f32x4 av = *(const f32x4*)(arow + 4 * k16);
+      const float* bk = bbase + (int64_t)16 * k16 * HEADC;
+      #pragma unroll
+      for (int n = 0; n < 2; ++n) {
+        f32x4 bv = *(const f32x4*)(bk + 64 * n);
+        #pragma unroll
+        for (int s = 0; s < 4; ++s)
+          acc[n] = __builtin_amdgcn_mfma_f32_16x16x4f32(av[s], bv[s], acc[n],
+                                                        0, 0, 0);
+      }
+    }
+    const int crow = (lane >> 4) * 4, ccol = lane & 15;
+    #pragma unroll
+    for (int n = 0; n < 2; ++n)
+      #pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        int col = n * 16 + ccol;
+        if (r0 + crow + i < p.B)
+          p.head[(r0 + crow + i) * HEADC + col] = acc[n][i] + p.bh[col];
+      }
+  }
+}
+
+}  // namespace
+
+extern "C" int rlsched_gru_seq_fwd(const int64_t* ip, const uint64_t* bufs,
+                                   void* stream) {
+  GruFwdArgs p;
+  p.T = ip[0]; p.Emb = ip[1]; p.D = ip[2]; p.H = ip[3];
+  if (p.H % 64) return -1;
+  const uint64_t* b = bufs;
+  p.obs = (const float*)b[0]; p.done = (const uint8_t*)b[1];
+  p.h0 = (const float*)b[2];
+  p.wtin = (const float*)b[3]; p.bin = (const float*)b[4];
+  p.wih = (const float*)b[5]; p.bih = (const float*)b[6];
+  p.whh = (const float*)b[7]; p.bhh = (const float*)b[8];
+  p.xs = (float*)b[9]; p.rs = (float*)b[10]; p.zs = (float*)b[11];
+  p.ns = (float*)b[12]; p.ghns = (float*)b[13];
+  p.hprevs = (float*)b[14]; p.hs = (float*)b[15];
+  const int Dp = ((int)p.D + 15) & ~15;
+  size_t lds = sizeof(float) * (ROWS * (Dp + 4) + 3 * ROWS * ((int)p.H + 4));
+  dim3 grid((unsigned)((p.Emb + ROWS - 1) / ROWS));
+  hipLaunchKernelGGL(gru_seq_fwd_kernel, grid, dim3(256), lds,
+                     (hipStream_t)stream, p);
+  return (int)hipGetLastError();
+}
+
+extern "C" int rlsched_gru_seq_bwd(const int64_t* ip, const uint64_t* bufs,
+                                   void* stream) {
+  GruBwdArgs p;
+  p.T = ip[0]; p.Emb = ip[1]; p.H = ip[2];
+  if (p.H % 64) return -1;
+  const uint64_t* b = bufs;
+  p.dh_ext = (const float*)b[0]; p.done = (const uint8_t*)b[1];
+  p.rs = (const float*)b[2]; p.zs = (const float*)b[3];
+  p.ns = (const float*)b[4]; p.ghns = (const float*)b[5];
+  p.hprevs = (const float*)b[6]; p.whh_bwd = (const float*)b[7];
+  p.dgi_all = (float*)b[8]; p.dgh_all = (float*)b[9];
+  size_t lds = sizeof(float) * (ROWS * (3 * (int)p.H + 4)
+                                + 2 * ROWS * ((int)p.H + 4));
+  dim3 grid((unsigned)((p.Emb + ROWS - 1) / ROWS));
+  hipLaunchKernelGGL(gru_seq_bwd_kernel, grid, dim3(256), lds,
+                     (hipStream_t)stream, p);
+  return (int)hipGetLastError();
+}
+
+extern "C" int rlsched_head_fwd(const int64_t* ip, const uint64_t* bufs,
+                                void* stream) {
+  HeadFwdArgs p;
+  p.B = ip[0]; p.H = ip[1];
+  if (p.H % 64) return -1;
+  p.h = (const float*)bufs[0];
+  p.wth = (const float*)bufs[1];
+  p.bh = (const float*)bufs[2];
+  p.head = (float*)bufs[3];
+  size_t lds = sizeof(float) * ROWS * ((int)p.H + 4);
+  dim3 grid((unsigned)((p.B + ROWS - 1) / ROWS));
+  hipLaunchKernelGGL(head_fwd_kernel, grid, dim3(256), lds,
+                     (hipStream_t)stream, p);
+  return (int)hipGetLastError();
+}

@@ -235,8 +235,12 @@ __global__ __launch_bounds__(256) void bwd_dx_kernel(BwdDxArgs p) {
         int col = c0 + n * 16 + ccol;
         int64_t row = row0 + crow + i;
         if (row < p.B) {
-          float hv = p.h[row * N + col];
-          p.dX[row * N + col] = acc[n][i] * (1.0f - hv * hv);
+          float g = acc[n][i];
+          if (p.h) {           // fused tanh' (null h -> plain GEMM)
+            float hv = p.h[row * N + col];
+            g *= (1.0f - hv * hv);
+          }
+          p.dX[row * N + col] = g;
         }
       }
   }

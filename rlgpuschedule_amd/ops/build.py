@@ -21,6 +21,7 @@ _LIBS = {
     "librlsched_env.so": ["env_step.hip"],
     "librlsched_nn.so": ["nn_kernels.hip"],
     "librlsched_train.so": ["train_kernels.hip"],
+    "librlsched_gru.so": ["gru_train.hip"],
 }
 
 

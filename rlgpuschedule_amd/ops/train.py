@@ -67,10 +67,20 @@ def ppo_head(head, mask, actions, old_logp, old_value, adv, ret,
 
 
 def bwd_dx(dY, wtp, h, dX):
-    """dX = (dY @ W^T) * (1 - h^2). dY [B,K], wtp packed [K,N], h/dX [B,N]."""
+    """dX = (dY @ W^T) [* (1 - h^2) if h is not None].
+    dY [B,K], wtp packed [K,N], dX [B,N]."""
     B, K = dY.shape
-    N = h.shape[1]
-    _call("rlsched_bwd_dx", [B, K, N], _ptrs(dY, wtp, h, dX), _stream(dY))
+    N = dX.shape[1]
+    hp = h.data_ptr() if h is not None else 0
+    ipa = np.asarray([B, K, N], np.int64)
+    bufs = np.array([dY.data_ptr(), wtp.data_ptr(), hp, dX.data_ptr()],
+                    dtype=np.uint64)
+    rc = load().rlsched_bwd_dx(
+        ipa.ctypes.data_as(ctypes.c_void_p),
+        bufs.ctypes.data_as(ctypes.c_void_p),
+        ctypes.c_void_p(_stream(dY)))
+    if rc != 0:
+        raise RuntimeError(f"rlsched_bwd_dx failed: {rc}")
 
 
 def bwd_dw(X, dY, dW, bc: int = 1024):

@@ -123,6 +123,9 @@ def train(cfg: TrainConfig, iterations: int | None = None,
     elif (on_gpu_fast and cfg.algo == "ppo" and not policy.is_recurrent):
         from .agents.ppo_hip import HipPPOLearner
         learner = HipPPOLearner(policy, cfg.ppo, grad_sync)
+    elif on_gpu_fast and cfg.algo == "ppo" and policy.is_recurrent:
+        from .agents.ppo_gru_hip import HipGRUPPOLearner
+        learner = HipGRUPPOLearner(policy, cfg.ppo, grad_sync)
     else:
         learner = make_learner(cfg, policy, grad_sync)
     start_it = 0
