@@ -31,7 +31,7 @@ from rlgpuschedule_amd.agents.ppo import PPOConfig, PPOLearner
 from rlgpuschedule_amd.agents.rollout import Collector, TorchEnvAdapter
 from rlgpuschedule_amd.env.gym_api import CpuVectorEnv
 from rlgpuschedule_amd.env.spec import cluster64
-from rlgpuschedule_amd.models.policy import MLPPolicy
+from rlgpuschedule_amd.models.policy import GRUPolicy, MLPPolicy
 from rlgpuschedule_amd.parallel import dist as pdist
 
 import torch.distributed as tdist
@@ -45,6 +45,7 @@ def main():
     ap.add_argument("--envs", type=int, default=8192, help="envs per GPU")
     ap.add_argument("--rollout", type=int, default=64)
     ap.add_argument("--hidden", type=int, default=256)
+    ap.add_argument("--policy", choices=["mlp", "gru"], default="mlp")
     ap.add_argument("--minibatches", type=int, default=4)
     ap.add_argument("--shards", type=int, default=2,
                     help="env shards on separate HIP streams (overlaps "
@@ -76,7 +77,8 @@ def main():
             CpuVectorEnv(spec, E, trace_kind="philly", jobs_per_env=64,
                          seed=97 + rank), device)
 
-    policy = MLPPolicy(spec.obs_dim, spec.num_actions, args.hidden).to(device)
+    pcls = GRUPolicy if args.policy == "gru" else MLPPolicy
+    policy = pcls(spec.obs_dim, spec.num_actions, args.hidden).to(device)
     pdist.broadcast_params(policy)
     grad_sync = pdist.FlatGradAllreduce(policy) if world > 1 else None
     # Throughput-oriented PPO: 1 epoch x 8 minibatches (documented in config)
@@ -84,8 +86,12 @@ def main():
     if has_gpu:
         from rlgpuschedule_amd.agents.ppo_gpu import FusedPPOLearner
         from rlgpuschedule_amd.agents.ppo_hip import HipPPOLearner
+        from rlgpuschedule_amd.agents.ppo_gru_hip import HipGRUPPOLearner
         from rlgpuschedule_amd.agents.rollout_gpu import GpuRolloutEngine
-        lcls = HipPPOLearner if args.learner == "hip" else FusedPPOLearner
+        if policy.is_recurrent:
+            lcls = HipGRUPPOLearner
+        else:
+            lcls = HipPPOLearner if args.learner == "hip" else FusedPPOLearner
         learner = lcls(policy, cfg, grad_sync)
         collector = GpuRolloutEngine(env, policy, T, seed=555 + rank)
     else:
@@ -143,7 +149,8 @@ def main():
             "mean_jct_us": round(st.get("mean_jct_us", 0.0), 1),
             "episodes": st.get("episodes", 0),
             "config": {
-                "model": "ppo_mlp_256x2",
+                "model": ("ppo_gru_256" if args.policy == "gru"
+                          else "ppo_mlp_256x2"),
                 "sim": "philly-like 64-GPU cluster (8 nodes x 8)",
                 "global_batch": T * E * n_gpus,
                 "seq_len": T,
