@@ -46,6 +46,7 @@ def main():
     ap.add_argument("--rollout", type=int, default=64)
     ap.add_argument("--hidden", type=int, default=256)
     ap.add_argument("--policy", choices=["mlp", "gru"], default="mlp")
+    ap.add_argument("--actor-dtype", choices=["fp32", "bf16"], default="fp32")
     ap.add_argument("--minibatches", type=int, default=4)
     ap.add_argument("--shards", type=int, default=2,
                     help="env shards on separate HIP streams (overlaps "
@@ -93,7 +94,8 @@ def main():
         else:
             lcls = HipPPOLearner if args.learner == "hip" else FusedPPOLearner
         learner = lcls(policy, cfg, grad_sync)
-        collector = GpuRolloutEngine(env, policy, T, seed=555 + rank)
+        collector = GpuRolloutEngine(env, policy, T, seed=555 + rank,
+                                     actor_dtype=args.actor_dtype)
     else:
         learner = PPOLearner(policy, cfg, grad_sync)
         collector = Collector(env, policy, T, device, seed=555 + rank)
@@ -144,7 +146,8 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,  # no published reference numbers
-            "dtype": "fp32",
+            "dtype": ("mixed (bf16 actor, fp32 update)"
+                      if args.actor_dtype == "bf16" else "fp32"),
             "data": "synthetic",
             "mean_jct_us": round(st.get("mean_jct_us", 0.0), 1),
             "episodes": st.get("episodes", 0),

@@ -13,13 +13,15 @@ from __future__ import annotations
 
 import torch
 
-from ..models.hip_actor import HipActorGRU, HipActorMLP
+from ..models.hip_actor import (HipActorGRU, HipActorMLP,
+                                HipActorMLPBf16)
 from .rollout import RolloutBuffer
 
 
 class GpuRolloutEngine:
     def __init__(self, env, policy, T: int, seed: int = 0,
-                 use_graph: bool = True, num_bufs: int = 1):
+                 use_graph: bool = True, num_bufs: int = 1,
+                 actor_dtype: str = "fp32"):
         """`env` may be a LIST of HipVectorEnvs (shards): each shard rolls
         out on its own HIP stream so shard A's env_step overlaps shard B's
         actor kernel (the per-step chain is serial within a shard). The
@@ -34,7 +36,10 @@ class GpuRolloutEngine:
         # one behavior-policy mirror per buffer: lets an async learner
         # refresh mirror i (after its update) while a rollout on the other
         # parity is still reading mirror i^1 (agents/impala.py)
-        acls = HipActorGRU if policy.is_recurrent else HipActorMLP
+        if policy.is_recurrent:
+            acls = HipActorGRU
+        else:
+            acls = HipActorMLPBf16 if actor_dtype == "bf16" else HipActorMLP
         self.actors = [acls(policy) for _ in range(num_bufs)]
         self.actor = self.actors[0]
         E = sum(e.num_envs for e in self.envs)

@@ -659,3 +659,187 @@ extern "C" int rlsched_gru_actor(const int64_t* ip, const uint64_t* bufs,
                      (hipStream_t)stream, p);
   return (int)hipGetLastError();
 }
+
+// ------------------------------------------------- bf16 fused MLP actor
+// Same structure as actor_mlp_kernel but the GEMMs run on
+// mfma_f32_16x16x32_bf16 (~13x the f32 MFMA rate; the actor is
+// MFMA-issue-bound per PMC). Weights/activations are bf16, accumulation
+// and the head/sampling stay f32. Fragment k-map: both operands use the
+// CONTIGUOUS per-lane map k = kk*8 + j (any bijective map shared by A and
+// B yields the correct dot product — verified by tools/probe_bf16_mfma).
+// B packed by models/hip_actor.py: wt[K,N].view(K/32,4(kk),8(j),N)
+// .permute(0,1,3,2) -> lane (kk,bc) reads ONE bf16x8 per 32-deep K step.
+namespace {
+
+typedef __bf16 bf16;
+typedef bf16 bf16x8 __attribute__((ext_vector_type(8)));
+
+struct ActorBf16Args {
+  int64_t E, D, H, A;
+  int64_t value_only;
+  const float* obs;
+  const uint8_t* mask;
+  const float* u;
+  const bf16* wt1;      // packed [Dp,H] bf16
+  const float* b1;
+  const bf16* wt2;      // packed [H,H]
+  const float* b2;
+  const bf16* wth;      // packed [H,32]
+  const float* bh;
+  int64_t* actions;
+  float* logp;
+  float* value;
+};
+
+// C[16 x 64] for one wave: A bf16 LDS tile (PLAIN row-major, contiguous k),
+// B packed global bf16. acc stays f32.
+__device__ __forceinline__ void mfma_core_bf16_16x64(
+    const bf16* __restrict__ a_lds, int lda, int K,
+    const bf16* __restrict__ b_pk, int ldb, int col0, int lane,
+    f32x4 acc[4]) {
+  const int ar = lane & 15;
+  const int kk = lane >> 4;
+  const int bc = lane & 15;
+  const bf16* arow = a_lds + ar * lda + kk * 8;
+  const bf16* bbase = b_pk + (int64_t)8 * (kk * ldb + col0 + bc);
+  for (int k32 = 0; k32 < (K >> 5); ++k32) {
+    bf16x8 av = *(const bf16x8*)(arow + 32 * k32);
+    const bf16* bk = bbase + (int64_t)32 * k32 * ldb;
+    #pragma unroll
+    for (int n = 0; n < 4; ++n) {
+      bf16x8 bv = *(const bf16x8*)(bk + 128 * n);
+      acc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(av, bv, acc[n],
+                                                       0, 0, 0);
+    }
+  }
+}
+
+__global__ __launch_bounds__(256) void actor_bf16_kernel(ActorBf16Args p) {
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int w = tid >> 6;
+  const int64_t row0 = (int64_t)blockIdx.x * ROWS;
+  const int D = (int)p.D, H = (int)p.H, A = (int)p.A;
+  const int Dp = (D + 31) & ~31;
+  // bf16 row strides: even (16B-alignable) and stride/2 dwords != 0 mod 64
+  const int lda_o = Dp + 8, lda_h = H + 8;
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  bf16* sobs = (bf16*)smem;                    // [ROWS][Dp+8]
+  bf16* h1 = sobs + ROWS * lda_o;              // [ROWS][H+8]
+  bf16* h2 = h1 + ROWS * lda_h;                // [ROWS][H+8]
+  float* shead = (float*)(h2 + ROWS * lda_h);  // [ROWS][33] f32
+
+  for (int ix = tid; ix < ROWS * Dp; ix += 256) {
+    int r = ix / Dp, d = ix % Dp;
+    float vv = 0.0f;
+    if (d < D && row0 + r < p.E) vv = p.obs[(row0 + r) * D + d];
+    sobs[r * lda_o + d] = (bf16)vv;
+  }
+  __syncthreads();
+  for (int c0 = w * 64; c0 < H; c0 += 256) {
+    f32x4 acc[4] = {};
+    mfma_core_bf16_16x64(sobs, lda_o, Dp, p.wt1, H, c0, lane, acc);
+    const int crow = (lane >> 4) * 4, ccol = lane & 15;
+    #pragma unroll
+    for (int n = 0; n < 4; ++n)
+      #pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        int col = c0 + n * 16 + ccol;
+        h1[(crow + i) * lda_h + col] = (bf16)tanhf(acc[n][i] + p.b1[col]);
+      }
+  }
+  __syncthreads();
+  for (int c0 = w * 64; c0 < H; c0 += 256) {
+    f32x4 acc[4] = {};
+    mfma_core_bf16_16x64(h1, lda_h, H, p.wt2, H, c0, lane, acc);
+    const int crow = (lane >> 4) * 4, ccol = lane & 15;
+    #pragma unroll
+    for (int n = 0; n < 4; ++n)
+      #pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        int col = c0 + n * 16 + ccol;
+        h2[(crow + i) * lda_h + col] = (bf16)tanhf(acc[n][i] + p.b2[col]);
+      }
+  }
+  __syncthreads();
+  if (w == 0) {
+    f32x4 acc[2] = {};
+    const int ar = lane & 15, kk = lane >> 4, bc = lane & 15;
+    const bf16* arow = h2 + ar * lda_h + kk * 8;
+    const bf16* bbase = p.wth + (int64_t)8 * (kk * HEADC + bc);
+    for (int k32 = 0; k32 < (H >> 5); ++k32) {
+      bf16x8 av = *(const bf16x8*)(arow + 32 * k32);
+      const bf16* bk = bbase + (int64_t)32 * k32 * HEADC;
+      #pragma unroll
+      for (int n = 0; n < 2; ++n) {
+        bf16x8 bv = *(const bf16x8*)(bk + 128 * n);
+        acc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(av, bv, acc[n],
+                                                         0, 0, 0);
+      }
+    }
+    const int crow = (lane >> 4) * 4, ccol = lane & 15;
+    #pragma unroll
+    for (int n = 0; n < 2; ++n)
+      #pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        int col = n * 16 + ccol;
+        shead[(crow + i) * (HEADC + 1) + col] = acc[n][i] + p.bh[col];
+      }
+  }
+  __syncthreads();
+  if (w == 0 && lane < ROWS) {
+    int64_t row = row0 + lane;
+    if (row < p.E) {
+      const float* hrow = shead + lane * (HEADC + 1);
+      p.value[row] = hrow[A];
+      if (!p.value_only) {
+        const uint8_t* mrow = p.mask + row * A;
+        float mx = -3.0e38f;
+        for (int i = 0; i < A; ++i)
+          if (mrow[i]) mx = fmaxf(mx, hrow[i]);
+        float sum = 0.0f;
+        for (int i = 0; i < A; ++i)
+          if (mrow[i]) sum += __expf(hrow[i] - mx);
+        float thresh = p.u[row] * sum;
+        float s = 0.0f;
+        int act = -1;
+        for (int i = 0; i < A; ++i) {
+          if (!mrow[i]) continue;
+          s += __expf(hrow[i] - mx);
+          act = i;
+          if (s >= thresh) break;
+        }
+        p.actions[row] = act;
+        p.logp[row] = (hrow[act] - mx) - __logf(sum);
+      }
+    }
+  }
+}
+
+}  // namespace
+
+extern "C" int rlsched_actor_bf16(const int64_t* ip, const uint64_t* bufs,
+                                  void* stream) {
+  ActorBf16Args p;
+  p.E = ip[0]; p.D = ip[1]; p.H = ip[2]; p.A = ip[3]; p.value_only = ip[4];
+  if (p.H % 64 || p.H % 32 || p.A + 1 > HEADC) return -1;
+  p.obs = (const float*)bufs[0];
+  p.mask = (const uint8_t*)bufs[1];
+  p.u = (const float*)bufs[2];
+  p.wt1 = (const bf16*)bufs[3];
+  p.b1 = (const float*)bufs[4];
+  p.wt2 = (const bf16*)bufs[5];
+  p.b2 = (const float*)bufs[6];
+  p.wth = (const bf16*)bufs[7];
+  p.bh = (const float*)bufs[8];
+  p.actions = (int64_t*)bufs[9];
+  p.logp = (float*)bufs[10];
+  p.value = (float*)bufs[11];
+  const int Dp = ((int)p.D + 31) & ~31;
+  size_t lds = sizeof(bf16) * (ROWS * (Dp + 8) + 2 * ROWS * ((int)p.H + 8))
+             + sizeof(float) * ROWS * (HEADC + 1);
+  dim3 grid((unsigned)((p.E + ROWS - 1) / ROWS));
+  hipLaunchKernelGGL(actor_bf16_kernel, grid, dim3(256), lds,
+                     (hipStream_t)stream, p);
+  return (int)hipGetLastError();
+}

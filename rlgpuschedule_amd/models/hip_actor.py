@@ -132,3 +132,65 @@ class HipActorGRU:
         gru_actor(obs, mask, value_out, done_prev, hx, hx_scratch, self.wtin,
                   self.bin, self.wih, self.bih, self.whh, self.bhh, self.wth,
                   self.bh, value_out, value_out, value_out, value_only=True)
+
+
+class HipActorMLPBf16:
+    """bf16-MFMA variant of HipActorMLP (~13x the f32 MFMA rate; the actor
+    is MFMA-issue-bound). Weights bf16, accumulation/head/sampling f32.
+    Packing: wt[K,N].view(K/32, 4(kk), 8(j), N).permute(0,1,3,2) — the
+    contiguous per-lane k-map shared by A and B fragments."""
+
+    def __init__(self, policy):
+        assert not policy.is_recurrent and len(policy.trunk) == 2
+        self.policy = policy
+        D, H, A = policy.obs_dim, policy.hidden, policy.num_actions
+        assert H % 64 == 0 and A + 1 <= HEADC
+        dev = policy.pi.weight.device
+        self.D, self.H, self.A = D, H, A
+        self.Dp = (D + 31) & ~31
+        self.wt1 = torch.zeros(self.Dp // 32, 4, H, 8, device=dev,
+                               dtype=torch.bfloat16)
+        self.b1 = torch.empty(H, device=dev)
+        self.wt2 = torch.empty(H // 32, 4, H, 8, device=dev,
+                               dtype=torch.bfloat16)
+        self.b2 = torch.empty(H, device=dev)
+        self.wth = torch.zeros(H // 32, 4, HEADC, 8, device=dev,
+                               dtype=torch.bfloat16)
+        self.bh = torch.zeros(HEADC, device=dev)
+        self.refresh()
+
+    @staticmethod
+    def _pack8(wt: torch.Tensor) -> torch.Tensor:
+        K, N = wt.shape
+        return wt.view(K // 32, 4, 8, N).permute(0, 1, 3, 2)
+
+    @torch.no_grad()
+    def refresh(self):
+        p = self.policy
+        D, H, A = self.D, self.H, self.A
+        if not hasattr(self, "_wt1_full"):
+            self._wt1_full = self.b1.new_zeros(self.Dp, H)
+            self._wth_full = self.b1.new_zeros(H, HEADC)
+        self._wt1_full[:D].copy_(p.trunk[0].weight.t())
+        self.wt1.copy_(self._pack8(self._wt1_full))
+        self.b1.copy_(p.trunk[0].bias)
+        self.wt2.copy_(self._pack8(p.trunk[1].weight.t()))
+        self.b2.copy_(p.trunk[1].bias)
+        self._wth_full[:, :A].copy_(p.pi.weight.t())
+        self._wth_full[:, A].copy_(p.v.weight[0])
+        self.wth.copy_(self._pack8(self._wth_full))
+        self.bh[:A].copy_(p.pi.bias)
+        self.bh[A] = p.v.bias[0]
+
+    def act(self, obs, mask, u, actions, logp, value):
+        from ..ops.nn import actor_bf16
+        actor_bf16(obs, mask, u, self.wt1, self.b1, self.wt2, self.b2,
+                   self.wth, self.bh, actions, logp, value)
+        return actions, logp, value
+
+    def values(self, obs, value_out, mask):
+        from ..ops.nn import actor_bf16
+        actor_bf16(obs, mask, value_out, self.wt1, self.b1, self.wt2,
+                   self.b2, self.wth, self.bh, value_out, value_out,
+                   value_out, value_only=True)
+        return value_out

@@ -153,3 +153,27 @@ def gru_actor(obs, mask, u, done_prev, hx, hx_out, wtin, bin_, wih, bih,
         bufs.ctypes.data_as(ctypes.c_void_p), ctypes.c_void_p(_stream(obs)))
     if rc != 0:
         raise RuntimeError(f"rlsched_gru_actor failed: {rc}")
+
+
+def _load_bf16():
+    lib = load()
+    if not hasattr(lib, "_bf16_ready"):
+        lib.rlsched_actor_bf16.restype = ctypes.c_int
+        lib.rlsched_actor_bf16.argtypes = [ctypes.c_void_p] * 3
+        lib._bf16_ready = True
+    return lib
+
+
+def actor_bf16(obs, mask, u, wt1, b1, wt2, b2, wth, bh, actions, logp, value,
+               value_only: bool = False):
+    """bf16-MFMA fused MLP actor (f32 accumulate/head/sampling)."""
+    E, D = obs.shape
+    H = wt2.shape[2]  # packed [H/32, 4, H, 8]
+    A = mask.shape[1]
+    ip = np.array([E, D, H, A, 1 if value_only else 0], np.int64)
+    bufs = _ptrs(obs, mask, u, wt1, b1, wt2, b2, wth, bh, actions, logp, value)
+    rc = _load_bf16().rlsched_actor_bf16(
+        ip.ctypes.data_as(ctypes.c_void_p),
+        bufs.ctypes.data_as(ctypes.c_void_p), ctypes.c_void_p(_stream(obs)))
+    if rc != 0:
+        raise RuntimeError(f"rlsched_actor_bf16 failed: {rc}")

@@ -140,3 +140,33 @@ def test_sharded_rollout_engine():
     assert torch.isfinite(buf.logp).all()
     for e in envs:
         e.validate_state()
+
+
+@pytest.mark.gpu
+def test_actor_bf16_matches_torch_policy():
+    from rlgpuschedule_amd.models.hip_actor import HipActorMLPBf16
+    from rlgpuschedule_amd.models.policy import MLPPolicy, sample_categorical
+    torch.manual_seed(4)
+    spec = cluster64()
+    D, A, H = spec.obs_dim, spec.num_actions, 256
+    E = 1000
+    policy = MLPPolicy(D, A, H).cuda()
+    obs = torch.randn(E, D, device="cuda")
+    mask = (torch.rand(E, A, device="cuda") < 0.5).to(torch.uint8)
+    mask[:, spec.noop_action] = 1
+    u = torch.rand(E, device="cuda")
+    actor = HipActorMLPBf16(policy)
+    actions = torch.empty(E, dtype=torch.int64, device="cuda")
+    logp = torch.empty(E, device="cuda")
+    value = torch.empty(E, device="cuda")
+    actor.act(obs, mask, u, actions, logp, value)
+    with torch.no_grad():
+        logits, v_ref, _ = policy(obs, mask, None)
+        a_ref, lp_ref, _ = sample_categorical(logits, u)
+    # bf16 weights/activations: loose tolerances, high action agreement
+    torch.testing.assert_close(value, v_ref, atol=5e-2, rtol=5e-2)
+    agree = (actions == a_ref)
+    assert agree.float().mean().item() > 0.97, agree.float().mean().item()
+    chosen = mask.gather(1, actions.unsqueeze(1))
+    assert (chosen == 1).all()
+    assert torch.isfinite(logp).all()
