@@ -304,3 +304,21 @@ class ClusterEnvRef:
             "jct_sum_us": self.jct_sum_us,
             "decisions": self.decisions,
         }
+
+    def validate_invariants(self) -> None:
+        """Accounting invariants (mirrors HipVectorEnv.validate_state)."""
+        s = self.spec
+        assert (self.node_free >= 0).all(), "negative node_free"
+        assert (self.node_free <= s.gpus_per_node).all(), "node_free > G"
+        assert self.run_count == int(self.run_valid.sum()), "run_count drift"
+        assert 0 <= self.pend_count <= s.queue_cap
+        used = sum(c for r in range(s.run_cap) if self.run_valid[r]
+                   for _, c in self.run_alloc[r])
+        assert used + int(self.node_free.sum()) == s.total_gpus, "GPU leak"
+        galloc = sum(int(self.run_gpus[r]) for r in range(s.run_cap)
+                     if self.run_valid[r])
+        assert galloc == used, "alloc pairs != allocated gpus"
+        for r in range(s.run_cap):
+            if self.run_valid[r]:
+                assert self.run_finish[r] > self.clock or \
+                    self.run_finish[r] >= self.clock, "stale running job"

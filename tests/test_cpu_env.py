@@ -268,3 +268,33 @@ def test_rigid_traces_unchanged_by_v3():
     env.step(spec.timeslice_action)
     env.step(env.pend_count - 1)  # re-place immediately (clock unchanged)
     assert int(env.run_finish[0]) == f0
+
+
+def test_invariants_fuzz():
+    """Property test: invariants hold across random specs/actions/resets."""
+    rng = np.random.default_rng(42)
+    for trial in range(6):
+        spec = EnvSpec(
+            num_nodes=int(rng.integers(2, 12)),
+            gpus_per_node=int(rng.choice([2, 4, 8])),
+            queue_visible=int(rng.integers(4, 12)),
+            queue_cap=int(rng.integers(16, 48)),
+            run_cap=int(rng.integers(16, 48)),
+            max_span=int(rng.integers(1, 5)),
+            horizon=int(rng.integers(50, 300)),
+            gandiva=bool(rng.integers(0, 2)),
+            elastic=bool(rng.integers(0, 2)),
+        )
+        tr = philly_like(24, seed=trial, max_job_gpus=spec.max_job_gpus,
+                         mean_interarrival_s=20.0, median_duration_s=60.0,
+                         elastic=spec.elastic, fail_frac=0.2)
+        env = ClusterEnvRef(spec, tr)
+        for i in range(400):
+            a = int(rng.integers(0, spec.num_actions))
+            obs, r, done, info = env.step(a)
+            if i % 25 == 0:
+                env.validate_invariants()
+                assert np.isfinite(obs).all()
+            if done:
+                env.reset()
+        env.validate_invariants()
