@@ -46,7 +46,7 @@ def main():
     ap.add_argument("--rollout", type=int, default=64)
     ap.add_argument("--hidden", type=int, default=256)
     ap.add_argument("--policy", choices=["mlp", "gru"], default="mlp")
-    ap.add_argument("--actor-dtype", choices=["fp32", "bf16"], default="fp32")
+    ap.add_argument("--actor-dtype", choices=["fp32", "bf16"], default="bf16")
     ap.add_argument("--minibatches", type=int, default=4)
     ap.add_argument("--shards", type=int, default=2,
                     help="env shards on separate HIP streams (overlaps "
