@@ -25,6 +25,8 @@ struct RlsParams {
   int64_t horizon;
   int64_t gandiva;    // 0/1
   int64_t elastic;    // 0/1 (ENV_SPEC v3)
+  int64_t debug_skip; // ablation bitmask (bench only): 1=slot obs/fits,
+                      // 2=whole obs/mask phase, 4=obs[3] next_event
 };
 
 // Device buffers; all row-major with env as the leading dim.

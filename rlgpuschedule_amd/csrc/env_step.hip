@@ -343,11 +343,13 @@ __device__ bool do_timeslice(const EnvView& v, Sc& s) {
 __device__ void write_obs_mask(const EnvView& v, const Sc& s, int lane) {
   const int K = (int)v.p->K, N = (int)v.p->N, A = (int)v.p->A;
   const int OD = (int)v.p->obs_dim;
+  const int64_t skip = v.p->debug_skip;
+  if (skip & 2) return;   // ablation only (bench/abl_env.py)
   for (int i = lane; i < OD; i += RLS_WAVE) v.obs[i] = 0.0f;
   int ft = 0;
   for (int n = lane; n < N; n += RLS_WAVE) ft += v.node_free[n];
   ft = wave_sum_i32(ft);
-  int64_t t_next = next_event(v, s, lane);
+  int64_t t_next = (skip & 4) ? 0 : next_event(v, s, lane);
   if (lane == 0) {
     v.obs[0] = (float)ft / (float)(v.p->N * v.p->G);
     v.obs[1] = (float)s.pend_count / (float)v.p->P;
@@ -355,7 +357,7 @@ __device__ void write_obs_mask(const EnvView& v, const Sc& s, int lane) {
     v.obs[3] = t_next != RLS_NO_EVENT ? lognorm(t_next - s.clock) : 0.0f;
   }
   // per-slot features + mask: lane k handles slot k (divergent fits scans)
-  if (lane < K) {
+  if (lane < K && !(skip & 1)) {
     bool valid = lane < s.pend_count;
     bool fits = false;
     if (valid) {
@@ -454,6 +456,7 @@ void unpack(const int64_t* params, const uint64_t* bufs, RlsParams& p, RlsBufs& 
   p.P = params[4]; p.R = params[5]; p.S = params[6]; p.A = params[7];
   p.obs_dim = params[8]; p.idle_tick_us = params[9]; p.time_scale_us = params[10];
   p.horizon = params[11]; p.gandiva = params[12]; p.elastic = params[13];
+  p.debug_skip = params[14];
   uint64_t* dst = reinterpret_cast<uint64_t*>(&b);
   for (int i = 0; i < RLS_NUM_BUFS; ++i) dst[i] = bufs[i];
 }

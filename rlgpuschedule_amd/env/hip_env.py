@@ -22,7 +22,8 @@ class HipVectorEnv:
                  jobs_per_env: int = 128, seed: int = 0,
                  device: str = "cuda", pool: JobTrace | None = None,
                  job_lo: np.ndarray | None = None,
-                 job_hi: np.ndarray | None = None, **trace_kw):
+                 job_hi: np.ndarray | None = None, debug_skip: int = 0,
+                 **trace_kw):
         if not torch.cuda.is_available():
             raise RuntimeError("HipVectorEnv needs a GPU (use CpuVectorEnv here)")
         assert spec.queue_visible <= 63, "kernel assumes K <= 63 (one lane/slot)"
@@ -79,6 +80,7 @@ class HipVectorEnv:
             spec.idle_tick_us, spec.time_scale_us, spec.horizon,
             1 if spec.gandiva else 0,
             1 if spec.elastic else 0,
+            int(debug_skip),
         ], dtype=np.int64)
         self.binding = EnvStepBinding(
             params, {k: v.data_ptr() for k, v in self.b.items()})
