@@ -37,8 +37,38 @@ def run(skip, E=8192, steps=300):
     return (time.perf_counter() - t0) / steps * 1e6
 
 
+def run_env(env, acts, steps=200):
+    import time
+    for _ in range(20):
+        env.step(acts)
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(steps):
+        env.step(acts)
+    torch.cuda.synchronize()
+    return (time.perf_counter() - t0) / steps * 1e6
+
+
 if __name__ == "__main__":
-    for skip, name in ((0, "full"), (1, "no slot obs/fits"),
-                       (4, "no next_event obs"), (5, "neither"),
-                       (2, "no obs/mask at all")):
-        print(f"skip={skip} ({name}): {run(skip):8.1f} us/step")
+    # within-probe interleaved A/B: shared envs, repeated rounds
+    E = 8192
+    spec = cluster64(horizon=1_000_000)
+    variants = [(0, "full"), (1, "no slot obs/fits"), (4, "no next_event"),
+                (5, "neither"), (2, "no obs/mask")]
+    envs = {s: HipVectorEnv(spec, E, trace_kind="philly", jobs_per_env=256,
+                            seed=1, debug_skip=s) for s, _ in variants}
+    g = torch.Generator(device="cuda").manual_seed(0)
+    acts = torch.where(torch.rand(E, device="cuda", generator=g) < 0.5,
+                       torch.zeros(E, dtype=torch.int64, device="cuda"),
+                       torch.full((E,), spec.noop_action, dtype=torch.int64,
+                                  device="cuda"))
+    for e in envs.values():
+        e.reset()
+    import collections
+    res = collections.defaultdict(list)
+    for rnd in range(3):
+        for s, name in variants:
+            res[s].append(run_env(envs[s], acts))
+    for s, name in variants:
+        v = sorted(res[s])
+        print(f"skip={s} ({name}): median {v[1]:7.1f} us  all={[round(x,1) for x in v]}")

@@ -13,7 +13,7 @@ from __future__ import annotations
 
 import torch
 
-from ..models.hip_actor import (HipActorGRU, HipActorMLP,
+from ..models.hip_actor import (HipActorGRU, HipActorGRUBf16, HipActorMLP,
                                 HipActorMLPBf16)
 from .rollout import RolloutBuffer
 
@@ -37,7 +37,7 @@ class GpuRolloutEngine:
         # refresh mirror i (after its update) while a rollout on the other
         # parity is still reading mirror i^1 (agents/impala.py)
         if policy.is_recurrent:
-            acls = HipActorGRU
+            acls = HipActorGRUBf16 if actor_dtype == "bf16" else HipActorGRU
         else:
             acls = HipActorMLPBf16 if actor_dtype == "bf16" else HipActorMLP
         self.actors = [acls(policy) for _ in range(num_bufs)]

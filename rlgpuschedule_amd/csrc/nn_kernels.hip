@@ -843,3 +843,194 @@ extern "C" int rlsched_actor_bf16(const int64_t* ip, const uint64_t* bufs,
                      (hipStream_t)stream, p);
   return (int)hipGetLastError();
 }
+
+// ------------------------------------------------- bf16 fused GRU actor
+// bf16-MFMA variant of gru_actor_kernel (same contiguous shared k-map as
+// the bf16 MLP actor; f32 accumulate, gates, head and sampling).
+namespace {
+
+struct GruBf16Args {
+  int64_t E, D, H, A;
+  int64_t value_only;
+  const float* obs;
+  const uint8_t* mask;
+  const float* u;
+  const uint8_t* done_prev;
+  const float* hx;
+  float* hx_out;
+  const bf16* wtin;     // packed bf16 [Dp,H]
+  const float* bin;
+  const bf16* wih;      // packed [H,3H]
+  const float* bih;
+  const bf16* whh;      // packed [H,3H]
+  const float* bhh;
+  const bf16* wth;      // packed [H,32]
+  const float* bh;
+  int64_t* actions;
+  float* logp;
+  float* value;
+};
+
+__global__ __launch_bounds__(256) void gru_bf16_kernel(GruBf16Args p) {
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int w = tid >> 6;
+  const int64_t row0 = (int64_t)blockIdx.x * ROWS;
+  const int D = (int)p.D, H = (int)p.H, A = (int)p.A;
+  const int Dp = (D + 31) & ~31;
+  const int lda_o = Dp + 8, lda_h = H + 8;
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  bf16* sobs = (bf16*)smem;                  // [ROWS][Dp+8]
+  bf16* sx = sobs + ROWS * lda_o;            // [ROWS][H+8]
+  bf16* shx = sx + ROWS * lda_h;             // [ROWS][H+8]
+  bf16* shn = shx + ROWS * lda_h;            // [ROWS][H+8] (h')
+  float* shxf = (float*)(shn + ROWS * lda_h);  // [ROWS][H] f32 h~ (exact z*h)
+  float* shead = shxf + ROWS * H;            // [ROWS][33]
+
+  for (int ix = tid; ix < ROWS * Dp; ix += 256) {
+    int r = ix / Dp, d = ix % Dp;
+    float vv = 0.0f;
+    if (d < D && row0 + r < p.E) vv = p.obs[(row0 + r) * D + d];
+    sobs[r * lda_o + d] = (bf16)vv;
+  }
+  for (int ix = tid; ix < ROWS * H; ix += 256) {
+    int r = ix / H, d = ix % H;
+    float vv = 0.0f;
+    int64_t row = row0 + r;
+    if (row < p.E) {
+      vv = p.hx[row * H + d];
+      if (p.done_prev && p.done_prev[row]) vv = 0.0f;
+    }
+    shx[r * lda_h + d] = (bf16)vv;
+    shxf[r * H + d] = vv;
+  }
+  __syncthreads();
+  for (int c0 = w * 64; c0 < H; c0 += 256) {
+    f32x4 acc[4] = {};
+    mfma_core_bf16_16x64(sobs, lda_o, Dp, p.wtin, H, c0, lane, acc);
+    const int crow = (lane >> 4) * 4, ccol = lane & 15;
+    #pragma unroll
+    for (int n = 0; n < 4; ++n)
+      #pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        int col = c0 + n * 16 + ccol;
+        sx[(crow + i) * lda_h + col] = (bf16)tanhf(acc[n][i] + p.bin[col]);
+      }
+  }
+  __syncthreads();
+  for (int c0 = w * 64; c0 < H; c0 += 256) {
+    f32x4 gir[4] = {}, giz[4] = {}, gin[4] = {};
+    f32x4 ghr[4] = {}, ghz[4] = {}, ghn[4] = {};
+    mfma_core_bf16_16x64(sx, lda_h, H, p.wih, 3 * H, 0 * H + c0, lane, gir);
+    mfma_core_bf16_16x64(sx, lda_h, H, p.wih, 3 * H, 1 * H + c0, lane, giz);
+    mfma_core_bf16_16x64(sx, lda_h, H, p.wih, 3 * H, 2 * H + c0, lane, gin);
+    mfma_core_bf16_16x64(shx, lda_h, H, p.whh, 3 * H, 0 * H + c0, lane, ghr);
+    mfma_core_bf16_16x64(shx, lda_h, H, p.whh, 3 * H, 1 * H + c0, lane, ghz);
+    mfma_core_bf16_16x64(shx, lda_h, H, p.whh, 3 * H, 2 * H + c0, lane, ghn);
+    const int crow = (lane >> 4) * 4, ccol = lane & 15;
+    #pragma unroll
+    for (int n = 0; n < 4; ++n)
+      #pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        int col = c0 + n * 16 + ccol;
+        int r = crow + i;
+        float rr = sigmoidf_(gir[n][i] + p.bih[col]
+                             + ghr[n][i] + p.bhh[col]);
+        float zz = sigmoidf_(giz[n][i] + p.bih[H + col]
+                             + ghz[n][i] + p.bhh[H + col]);
+        float nn = tanhf(gin[n][i] + p.bih[2 * H + col]
+                         + rr * (ghn[n][i] + p.bhh[2 * H + col]));
+        float hold = shxf[r * H + col];   // f32 carry for the z*h path
+        float hnew = (1.0f - zz) * nn + zz * hold;
+        shn[r * lda_h + col] = (bf16)hnew;
+        int64_t row = row0 + r;
+        if (row < p.E) p.hx_out[row * H + col] = hnew;
+      }
+  }
+  __syncthreads();
+  if (w == 0) {
+    f32x4 acc[2] = {};
+    const int ar = lane & 15, kk = lane >> 4, bc = lane & 15;
+    const bf16* arow = shn + ar * lda_h + kk * 8;
+    const bf16* bbase = p.wth + (int64_t)8 * (kk * HEADC + bc);
+    for (int k32 = 0; k32 < (H >> 5); ++k32) {
+      bf16x8 av = *(const bf16x8*)(arow + 32 * k32);
+      const bf16* bk = bbase + (int64_t)32 * k32 * HEADC;
+      #pragma unroll
+      for (int n = 0; n < 2; ++n) {
+        bf16x8 bv = *(const bf16x8*)(bk + 128 * n);
+        acc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(av, bv, acc[n],
+                                                         0, 0, 0);
+      }
+    }
+    const int crow = (lane >> 4) * 4, ccol = lane & 15;
+    #pragma unroll
+    for (int n = 0; n < 2; ++n)
+      #pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        int col = n * 16 + ccol;
+        shead[(crow + i) * (HEADC + 1) + col] = acc[n][i] + p.bh[col];
+      }
+  }
+  __syncthreads();
+  if (w == 0 && lane < ROWS) {
+    int64_t row = row0 + lane;
+    if (row < p.E) {
+      const float* hrow = shead + lane * (HEADC + 1);
+      p.value[row] = hrow[A];
+      if (!p.value_only) {
+        const uint8_t* mrow = p.mask + row * A;
+        float mx = -3.0e38f;
+        for (int i = 0; i < A; ++i)
+          if (mrow[i]) mx = fmaxf(mx, hrow[i]);
+        float sum = 0.0f;
+        for (int i = 0; i < A; ++i)
+          if (mrow[i]) sum += __expf(hrow[i] - mx);
+        float thresh = p.u[row] * sum;
+        float s = 0.0f;
+        int act = -1;
+        for (int i = 0; i < A; ++i) {
+          if (!mrow[i]) continue;
+          s += __expf(hrow[i] - mx);
+          act = i;
+          if (s >= thresh) break;
+        }
+        p.actions[row] = act;
+        p.logp[row] = (hrow[act] - mx) - __logf(sum);
+      }
+    }
+  }
+}
+
+}  // namespace
+
+extern "C" int rlsched_gru_bf16(const int64_t* ip, const uint64_t* bufs,
+                                void* stream) {
+  GruBf16Args p;
+  p.E = ip[0]; p.D = ip[1]; p.H = ip[2]; p.A = ip[3]; p.value_only = ip[4];
+  if (p.H % 64 || p.A + 1 > HEADC) return -1;
+  p.obs = (const float*)bufs[0];
+  p.mask = (const uint8_t*)bufs[1];
+  p.u = (const float*)bufs[2];
+  p.done_prev = (const uint8_t*)bufs[3];
+  p.hx = (const float*)bufs[4];
+  p.hx_out = (float*)bufs[5];
+  p.wtin = (const bf16*)bufs[6];
+  p.bin = (const float*)bufs[7];
+  p.wih = (const bf16*)bufs[8];
+  p.bih = (const float*)bufs[9];
+  p.whh = (const bf16*)bufs[10];
+  p.bhh = (const float*)bufs[11];
+  p.wth = (const bf16*)bufs[12];
+  p.bh = (const float*)bufs[13];
+  p.actions = (int64_t*)bufs[14];
+  p.logp = (float*)bufs[15];
+  p.value = (float*)bufs[16];
+  const int Dp = ((int)p.D + 31) & ~31;
+  size_t lds = sizeof(bf16) * (ROWS * (Dp + 8) + 3 * ROWS * ((int)p.H + 8))
+             + sizeof(float) * (ROWS * (int)p.H + ROWS * (HEADC + 1));
+  dim3 grid((unsigned)((p.E + ROWS - 1) / ROWS));
+  hipLaunchKernelGGL(gru_bf16_kernel, grid, dim3(256), lds,
+                     (hipStream_t)stream, p);
+  return (int)hipGetLastError();
+}

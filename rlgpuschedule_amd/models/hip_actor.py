@@ -194,3 +194,60 @@ class HipActorMLPBf16:
                    self.b2, self.wth, self.bh, value_out, value_out,
                    value_out, value_only=True)
         return value_out
+
+
+class HipActorGRUBf16:
+    """bf16-MFMA variant of HipActorGRU (f32 accumulate/gates/sampling)."""
+
+    def __init__(self, policy):
+        assert policy.is_recurrent
+        D, H, A = policy.obs_dim, policy.hidden, policy.num_actions
+        assert H % 64 == 0 and A + 1 <= HEADC
+        dev = policy.pi.weight.device
+        self.policy = policy
+        self.D, self.H, self.A = D, H, A
+        self.Dp = (D + 31) & ~31
+        bf = torch.bfloat16
+        self.wtin = torch.zeros(self.Dp // 32, 4, H, 8, device=dev, dtype=bf)
+        self.bin = torch.empty(H, device=dev)
+        self.wih = torch.empty(H // 32, 4, 3 * H, 8, device=dev, dtype=bf)
+        self.bih = torch.empty(3 * H, device=dev)
+        self.whh = torch.empty(H // 32, 4, 3 * H, 8, device=dev, dtype=bf)
+        self.bhh = torch.empty(3 * H, device=dev)
+        self.wth = torch.zeros(H // 32, 4, HEADC, 8, device=dev, dtype=bf)
+        self.bh = torch.zeros(HEADC, device=dev)
+        self.refresh()
+
+    _pack8 = staticmethod(HipActorMLPBf16._pack8)
+
+    @torch.no_grad()
+    def refresh(self):
+        p = self.policy
+        D, H, A = self.D, self.H, self.A
+        if not hasattr(self, "_win_full"):
+            self._win_full = self.bin.new_zeros(self.Dp, H)
+            self._wth_full = self.bin.new_zeros(H, HEADC)
+        self._win_full[:D].copy_(p.inp.weight.t())
+        self.wtin.copy_(self._pack8(self._win_full))
+        self.bin.copy_(p.inp.bias)
+        self.wih.copy_(self._pack8(p.cell.weight_ih.t().contiguous()))
+        self.bih.copy_(p.cell.bias_ih)
+        self.whh.copy_(self._pack8(p.cell.weight_hh.t().contiguous()))
+        self.bhh.copy_(p.cell.bias_hh)
+        self._wth_full[:, :A].copy_(p.pi.weight.t())
+        self._wth_full[:, A].copy_(p.v.weight[0])
+        self.wth.copy_(self._pack8(self._wth_full))
+        self.bh[:A].copy_(p.pi.bias)
+        self.bh[A] = p.v.bias[0]
+
+    def act(self, obs, mask, u, done_prev, hx, hx_out, actions, logp, value):
+        from ..ops.nn import gru_bf16
+        gru_bf16(obs, mask, u, done_prev, hx, hx_out, self.wtin, self.bin,
+                 self.wih, self.bih, self.whh, self.bhh, self.wth, self.bh,
+                 actions, logp, value)
+
+    def values(self, obs, mask, done_prev, hx, hx_scratch, value_out):
+        from ..ops.nn import gru_bf16
+        gru_bf16(obs, mask, value_out, done_prev, hx, hx_scratch, self.wtin,
+                 self.bin, self.wih, self.bih, self.whh, self.bhh, self.wth,
+                 self.bh, value_out, value_out, value_out, value_only=True)

@@ -177,3 +177,30 @@ def actor_bf16(obs, mask, u, wt1, b1, wt2, b2, wth, bh, actions, logp, value,
         bufs.ctypes.data_as(ctypes.c_void_p), ctypes.c_void_p(_stream(obs)))
     if rc != 0:
         raise RuntimeError(f"rlsched_actor_bf16 failed: {rc}")
+
+
+def gru_bf16(obs, mask, u, done_prev, hx, hx_out, wtin, bin_, wih, bih,
+             whh, bhh, wth, bh, actions, logp, value,
+             value_only: bool = False):
+    """bf16-MFMA fused GRU actor step."""
+    lib = load()
+    if not hasattr(lib, "_grub_ready"):
+        lib.rlsched_gru_bf16.restype = ctypes.c_int
+        lib.rlsched_gru_bf16.argtypes = [ctypes.c_void_p] * 3
+        lib._grub_ready = True
+    E, D = obs.shape
+    H = hx.shape[1]
+    A = mask.shape[1]
+    ip = np.array([E, D, H, A, 1 if value_only else 0], np.int64)
+    dp = done_prev.data_ptr() if done_prev is not None else 0
+    bufs = np.array([obs.data_ptr(), mask.data_ptr(), u.data_ptr(), dp,
+                     hx.data_ptr(), hx_out.data_ptr(), wtin.data_ptr(),
+                     bin_.data_ptr(), wih.data_ptr(), bih.data_ptr(),
+                     whh.data_ptr(), bhh.data_ptr(), wth.data_ptr(),
+                     bh.data_ptr(), actions.data_ptr(), logp.data_ptr(),
+                     value.data_ptr()], dtype=np.uint64)
+    rc = lib.rlsched_gru_bf16(
+        ip.ctypes.data_as(ctypes.c_void_p),
+        bufs.ctypes.data_as(ctypes.c_void_p), ctypes.c_void_p(_stream(obs)))
+    if rc != 0:
+        raise RuntimeError(f"rlsched_gru_bf16 failed: {rc}")

@@ -68,3 +68,36 @@ def test_gru_rollout_engine_and_recurrent_ppo():
         chosen = buf.mask[t].gather(1, buf.actions[t].unsqueeze(1))
         assert (chosen == 1).all()
     assert torch.isfinite(buf.h0).all()
+
+
+@pytest.mark.gpu
+def test_gru_bf16_actor_close_to_torch():
+    from rlgpuschedule_amd.models.hip_actor import HipActorGRUBf16
+    from rlgpuschedule_amd.models.policy import GRUPolicy, sample_categorical
+    import torch
+    torch.manual_seed(7)
+    spec = cluster256()
+    D, A, H = spec.obs_dim, spec.num_actions, 256
+    E = 1000
+    policy = GRUPolicy(D, A, H).cuda()
+    obs = torch.randn(E, D, device="cuda")
+    mask = (torch.rand(E, A, device="cuda") < 0.5).to(torch.uint8)
+    mask[:, spec.noop_action] = 1
+    u = torch.rand(E, device="cuda")
+    hx = torch.randn(E, H, device="cuda") * 0.5
+    done_prev = (torch.rand(E, device="cuda") < 0.3).to(torch.uint8)
+    actor = HipActorGRUBf16(policy)
+    actions = torch.empty(E, dtype=torch.int64, device="cuda")
+    logp = torch.empty(E, device="cuda")
+    value = torch.empty(E, device="cuda")
+    hx_out = torch.empty(E, H, device="cuda")
+    actor.act(obs, mask, u, done_prev, hx, hx_out, actions, logp, value)
+    with torch.no_grad():
+        hx_ref = hx * (1.0 - done_prev.float()).unsqueeze(-1)
+        logits, v_ref, hx_new = policy(obs, mask, hx_ref)
+        a_ref, lp_ref, _ = sample_categorical(logits, u)
+    torch.testing.assert_close(hx_out, hx_new, atol=5e-2, rtol=5e-2)
+    torch.testing.assert_close(value, v_ref, atol=8e-2, rtol=8e-2)
+    assert (actions == a_ref).float().mean().item() > 0.95
+    chosen = mask.gather(1, actions.unsqueeze(1))
+    assert (chosen == 1).all()
