@@ -34,7 +34,6 @@ from rlgpuschedule_amd.env.spec import cluster64
 from rlgpuschedule_amd.models.policy import GRUPolicy, MLPPolicy
 from rlgpuschedule_amd.parallel import dist as pdist
 
-import torch.distributed as tdist
 
 
 def main():

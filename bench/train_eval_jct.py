@@ -5,7 +5,6 @@ Usage (GPU box): python bench/train_eval_jct.py [--iters 300]
 Writes results JSON to stdout; used for docs/PERF.md quality table.
 """
 import argparse
-import dataclasses
 import json
 import sys
 import time

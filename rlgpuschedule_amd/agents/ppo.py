@@ -111,7 +111,6 @@ class PPOLearner:
             perm = torch.randperm(B, device=obs.device)
             for i in range(cfg.minibatches):
                 ix = perm[i * mb:(i + 1) * mb]
-                from ..models.policy import masked_logits
                 logits, value, _ = self.policy(obs[ix], mask[ix], None)
                 logp, entropy = evaluate_actions(logits, acts[ix])
                 loss = self._loss(logp, entropy, value, old_logp[ix],

@@ -4,7 +4,7 @@ The five named configs mirror BASELINE.json's config list.
 """
 from __future__ import annotations
 
-from dataclasses import dataclass, field, replace
+from dataclasses import dataclass, field
 
 import torch
 
