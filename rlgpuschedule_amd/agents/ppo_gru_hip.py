@@ -30,11 +30,13 @@ from .rollout import RolloutBuffer
 
 class HipGRUPPOLearner:
     def __init__(self, policy, cfg: PPOConfig, grad_sync=None,
-                 use_graph: bool = True):
+                 use_graph: bool = True, use_hip_adam: bool = True):
         assert policy.is_recurrent
         self.policy = policy
         self.cfg = cfg
         self.grad_sync = grad_sync
+        self.use_hip_adam = use_hip_adam
+        self.hip_opt = None
         self.use_graph = use_graph
         self.opt = torch.optim.Adam(policy.parameters(), lr=cfg.lr, eps=1e-5,
                                     capturable=self.use_graph)
@@ -90,14 +92,41 @@ class HipGRUPPOLearner:
             off += n
             return t
         self._s["gflat"] = gflat
-        self._s["dWin"] = carve(D, H)
-        self._s["dbin"] = carve(H)
-        self._s["dWih"] = carve(H, 3 * H)
-        self._s["dbih"] = carve(3 * H)
-        self._s["dWhh"] = carve(H, 3 * H)
-        self._s["dbhh"] = carve(3 * H)
-        self._s["dWh"] = carve(H, HEADC)
-        self._s["dbh"] = carve(HEADC)
+        offs = {}
+        for name, shape in (("dWin", (D, H)), ("dbin", (H,)),
+                            ("dWih", (H, 3 * H)), ("dbih", (3 * H,)),
+                            ("dWhh", (H, 3 * H)), ("dbhh", (3 * H,)),
+                            ("dWh", (H, HEADC)), ("dbh", (HEADC,))):
+            offs[name] = off
+            self._s[name] = carve(*shape)
+        if self.use_hip_adam:
+            from ..ops.train import HipAdam
+            from ..parallel.dist import allreduce_flat
+            pol = self.policy
+            W = lambda name, R, C, c0, c1, p: dict(
+                goff=offs[name], R=R, C=C, c0=c0, c1=c1, param=p,
+                transposed=True)
+            Bs = lambda name, C, c0, c1, p: dict(
+                goff=offs[name], R=1, C=C, c0=c0, c1=c1, param=p,
+                transposed=False)
+            segs = [
+                W("dWin", D, H, 0, H, pol.inp.weight),
+                W("dWih", H, 3 * H, 0, 3 * H, pol.cell.weight_ih),
+                W("dWhh", H, 3 * H, 0, 3 * H, pol.cell.weight_hh),
+                W("dWh", H, HEADC, 0, A, pol.pi.weight),
+                W("dWh", H, HEADC, A, A + 1, pol.v.weight),
+                Bs("dbin", H, 0, H, pol.inp.bias),
+                Bs("dbih", 3 * H, 0, 3 * H, pol.cell.bias_ih),
+                Bs("dbhh", 3 * H, 0, 3 * H, pol.cell.bias_hh),
+                Bs("dbh", HEADC, 0, A, pol.pi.bias),
+                Bs("dbh", HEADC, A, A + 1, pol.v.bias),
+            ]
+            self.hip_opt = HipAdam(
+                gflat, segs, lr=cfg.lr, eps=1e-5,
+                max_norm=cfg.max_grad_norm,
+                grad_reduce=allreduce_flat if self.grad_sync is not None
+                else None)
+            self.opt = self.hip_opt
 
     def _refill_perm(self):
         for ep in range(self.cfg.epochs):
@@ -158,21 +187,24 @@ class HipGRUPPOLearner:
                 colsum(s["dgh"], s["dbhh"])
                 bwd_dw(s["obs"].view(B, self._D), s["dxpre"], s["dWin"])
                 colsum(s["dxpre"], s["dbin"])
-                pol.inp.weight.grad.copy_(s["dWin"].t())
-                pol.inp.bias.grad.copy_(s["dbin"])
-                pol.cell.weight_ih.grad.copy_(s["dWih"].t())
-                pol.cell.bias_ih.grad.copy_(s["dbih"])
-                pol.cell.weight_hh.grad.copy_(s["dWhh"].t())
-                pol.cell.bias_hh.grad.copy_(s["dbhh"])
-                pol.pi.weight.grad.copy_(s["dWh"][:, :A].t())
-                pol.pi.bias.grad.copy_(s["dbh"][:A])
-                pol.v.weight.grad.copy_(s["dWh"][:, A:A + 1].t())
-                pol.v.bias.grad.copy_(s["dbh"][A:A + 1])
-                if self.grad_sync is not None:
-                    self.grad_sync(self.policy)
-                torch.nn.utils.clip_grad_norm_(pol.parameters(),
-                                               cfg.max_grad_norm)
-                self.opt.step()
+                if self.hip_opt is not None:
+                    self.hip_opt.step()   # fused clip+Adam on gflat
+                else:
+                    pol.inp.weight.grad.copy_(s["dWin"].t())
+                    pol.inp.bias.grad.copy_(s["dbin"])
+                    pol.cell.weight_ih.grad.copy_(s["dWih"].t())
+                    pol.cell.bias_ih.grad.copy_(s["dbih"])
+                    pol.cell.weight_hh.grad.copy_(s["dWhh"].t())
+                    pol.cell.bias_hh.grad.copy_(s["dbhh"])
+                    pol.pi.weight.grad.copy_(s["dWh"][:, :A].t())
+                    pol.pi.bias.grad.copy_(s["dbh"][:A])
+                    pol.v.weight.grad.copy_(s["dWh"][:, A:A + 1].t())
+                    pol.v.bias.grad.copy_(s["dbh"][A:A + 1])
+                    if self.grad_sync is not None:
+                        self.grad_sync(self.policy)
+                    torch.nn.utils.clip_grad_norm_(pol.parameters(),
+                                                   cfg.max_grad_norm)
+                    self.opt.step()
                 mi.refresh()
 
     def _capture(self, buf: RolloutBuffer):

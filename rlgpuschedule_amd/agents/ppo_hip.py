@@ -29,11 +29,13 @@ from .rollout import RolloutBuffer
 
 class HipPPOLearner:
     def __init__(self, policy, cfg: PPOConfig, grad_sync=None,
-                 use_graph: bool = True):
+                 use_graph: bool = True, use_hip_adam: bool = True):
         assert not policy.is_recurrent
         self.policy = policy
         self.cfg = cfg
         self.grad_sync = grad_sync
+        self.use_hip_adam = use_hip_adam
+        self.hip_opt = None
         # graph capture is attempted even with DP grad sync (RCCL ops are
         # capturable on recent ROCm); on capture failure we fall back to
         # eager execution (update() try/except)
@@ -84,12 +86,38 @@ class HipPPOLearner:
             off += n
             return t
         self._s["gflat"] = gflat
-        self._s["dW1"] = carve(D, H)
-        self._s["dW2"] = carve(H, H)
-        self._s["dWh"] = carve(H, HEADC)
-        self._s["db1"] = carve(H)
-        self._s["db2"] = carve(H)
-        self._s["dbh"] = carve(HEADC)
+        offs = {}
+        for name, shape in (("dW1", (D, H)), ("dW2", (H, H)),
+                            ("dWh", (H, HEADC)), ("db1", (H,)),
+                            ("db2", (H,)), ("dbh", (HEADC,))):
+            offs[name] = off
+            self._s[name] = carve(*shape)
+        if self.use_hip_adam:
+            from ..ops.train import HipAdam
+            from ..parallel.dist import allreduce_flat
+            pol = self.policy
+            W = lambda name, R, C, c0, c1, p: dict(
+                goff=offs[name], R=R, C=C, c0=c0, c1=c1, param=p,
+                transposed=True)
+            Bs = lambda name, C, c0, c1, p: dict(
+                goff=offs[name], R=1, C=C, c0=c0, c1=c1, param=p,
+                transposed=False)
+            segs = [
+                W("dW1", D, H, 0, H, pol.trunk[0].weight),
+                W("dW2", H, H, 0, H, pol.trunk[1].weight),
+                W("dWh", H, HEADC, 0, A, pol.pi.weight),
+                W("dWh", H, HEADC, A, A + 1, pol.v.weight),
+                Bs("db1", H, 0, H, pol.trunk[0].bias),
+                Bs("db2", H, 0, H, pol.trunk[1].bias),
+                Bs("dbh", HEADC, 0, A, pol.pi.bias),
+                Bs("dbh", HEADC, A, A + 1, pol.v.bias),
+            ]
+            self.hip_opt = HipAdam(
+                gflat, segs, lr=cfg.lr, eps=1e-5,
+                max_norm=cfg.max_grad_norm,
+                grad_reduce=allreduce_flat if self.grad_sync is not None
+                else None)
+            self.opt = self.hip_opt
 
     def _refill_perm(self):
         for ep in range(self.cfg.epochs):
@@ -142,20 +170,23 @@ class HipPPOLearner:
                 colsum(s["dhead"], s["dbh"])
                 colsum(s["dh2"], s["db2"])
                 colsum(s["dh1"], s["db1"])
-                # assemble into torch param grads (layout: Linear [out,in])
-                pol.trunk[0].weight.grad.copy_(s["dW1"].t())
-                pol.trunk[0].bias.grad.copy_(s["db1"])
-                pol.trunk[1].weight.grad.copy_(s["dW2"].t())
-                pol.trunk[1].bias.grad.copy_(s["db2"])
-                pol.pi.weight.grad.copy_(s["dWh"][:, :A].t())
-                pol.pi.bias.grad.copy_(s["dbh"][:A])
-                pol.v.weight.grad.copy_(s["dWh"][:, A:A + 1].t())
-                pol.v.bias.grad.copy_(s["dbh"][A:A + 1])
-                if self.grad_sync is not None:
-                    self.grad_sync(self.policy)
-                torch.nn.utils.clip_grad_norm_(pol.parameters(),
-                                               cfg.max_grad_norm)
-                self.opt.step()
+                if self.hip_opt is not None:
+                    self.hip_opt.step()   # fused clip+Adam on gflat
+                else:
+                    # torch fallback: assemble into param grads
+                    pol.trunk[0].weight.grad.copy_(s["dW1"].t())
+                    pol.trunk[0].bias.grad.copy_(s["db1"])
+                    pol.trunk[1].weight.grad.copy_(s["dW2"].t())
+                    pol.trunk[1].bias.grad.copy_(s["db2"])
+                    pol.pi.weight.grad.copy_(s["dWh"][:, :A].t())
+                    pol.pi.bias.grad.copy_(s["dbh"][:A])
+                    pol.v.weight.grad.copy_(s["dWh"][:, A:A + 1].t())
+                    pol.v.bias.grad.copy_(s["dbh"][A:A + 1])
+                    if self.grad_sync is not None:
+                        self.grad_sync(self.policy)
+                    torch.nn.utils.clip_grad_norm_(pol.parameters(),
+                                                   cfg.max_grad_norm)
+                    self.opt.step()
                 mi.refresh()   # params changed; next minibatch needs them
 
     def _capture(self, buf: RolloutBuffer):

@@ -111,3 +111,85 @@ def gather_mb(ix, obs, mask, acts, olp, ov, adv, ret,
           _ptrs(ix, obs, mask, acts, olp, ov, adv, ret,
                 o_obs, o_mask, o_acts, o_olp, o_ov, o_adv, o_ret),
           _stream(obs))
+
+
+def _load_adam():
+    lib = load()
+    if not hasattr(lib, "_adam_ready"):
+        lib.rlsched_adam_bump.restype = ctypes.c_int
+        lib.rlsched_adam_bump.argtypes = [ctypes.c_void_p] * 4
+        lib.rlsched_adam_step.restype = ctypes.c_int
+        lib.rlsched_adam_step.argtypes = [ctypes.c_void_p] * 4
+        lib._adam_ready = True
+    return lib
+
+
+class HipAdam:
+    """Fused clip+Adam over the flat grad buffer (2 kernel launches).
+
+    segments: list of dicts with keys
+      goff (flat offset), R, C (full grad row stride), c0, c1 (column
+      slice), param (torch tensor, f32), transposed (grad [in,out] ->
+      param [out,in]; biases/slices use the compact layout).
+    Duck-types torch optimizers for checkpointing (state_dict).
+    """
+
+    def __init__(self, gflat, segments, lr=3e-4, betas=(0.9, 0.999),
+                 eps=1e-5, max_norm=0.5, grad_reduce=None):
+        self.gflat = gflat
+        self.lr, self.betas, self.eps = lr, betas, eps
+        self.max_norm = max_norm
+        self.grad_reduce = grad_reduce  # DP hook: all-reduce gflat
+        dev = gflat.device
+        total = sum(s["R"] * (s["c1"] - s["c0"]) for s in segments)
+        self.m = torch.zeros(total, device=dev)
+        self.v = torch.zeros(total, device=dev)
+        self.t = torch.zeros(1, device=dev)
+        self.bc = torch.ones(2, device=dev)
+        self.coef = torch.ones(1, device=dev)
+        ip = [len(segments)]
+        moff = 0
+        for s in segments:
+            ip += [s["goff"], moff, s["R"], s["C"], s["c0"], s["c1"]]
+            moff += s["R"] * (s["c1"] - s["c0"])
+        ip += [1 if s["transposed"] else 0 for s in segments]
+        self._ip = np.asarray(ip, np.int64)
+        self._bump_ip = np.zeros(1, np.int64)
+        self._bump_fp = np.asarray(betas, np.float64)
+        self._fp = np.asarray([lr, betas[0], betas[1], eps], np.float64)
+        self._bufs = np.array(
+            [gflat.data_ptr(), self.m.data_ptr(), self.v.data_ptr(),
+             self.coef.data_ptr(), self.bc.data_ptr()]
+            + [s["param"].data_ptr() for s in segments], dtype=np.uint64)
+        self._bump_bufs = np.array([self.t.data_ptr(), self.bc.data_ptr()],
+                                   dtype=np.uint64)
+
+    def step(self):
+        if self.grad_reduce is not None:
+            self.grad_reduce(self.gflat)
+        norm = torch.linalg.vector_norm(self.gflat)
+        torch.clamp(self.max_norm / (norm + 1e-6), max=1.0, out=self.coef)
+        lib = _load_adam()
+        st = _stream(self.gflat)
+        rc = lib.rlsched_adam_bump(
+            self._bump_ip.ctypes.data_as(ctypes.c_void_p),
+            self._bump_fp.ctypes.data_as(ctypes.c_void_p),
+            self._bump_bufs.ctypes.data_as(ctypes.c_void_p),
+            ctypes.c_void_p(st))
+        if rc == 0:
+            rc = lib.rlsched_adam_step(
+                self._ip.ctypes.data_as(ctypes.c_void_p),
+                self._fp.ctypes.data_as(ctypes.c_void_p),
+                self._bufs.ctypes.data_as(ctypes.c_void_p),
+                ctypes.c_void_p(st))
+        if rc != 0:
+            raise RuntimeError(f"hip adam failed: {rc}")
+
+    # torch-optimizer duck typing for ckpt/io.py
+    def state_dict(self):
+        return {"hip_adam": True, "m": self.m, "v": self.v, "t": self.t}
+
+    def load_state_dict(self, sd):
+        self.m.copy_(sd["m"].to(self.m.device))
+        self.v.copy_(sd["v"].to(self.v.device))
+        self.t.copy_(sd["t"].to(self.t.device))

@@ -85,6 +85,13 @@ class FlatGradAllreduce:
             off += n
 
 
+def allreduce_flat(t: torch.Tensor) -> None:
+    """Mean all-reduce of a flat gradient tensor (HipAdam DP hook)."""
+    if dist.is_initialized() and dist.get_world_size() > 1:
+        t.div_(dist.get_world_size())
+        dist.all_reduce(t)
+
+
 def allreduce_scalar(x: float, op: str = "max") -> float:
     if not dist.is_initialized():
         return x

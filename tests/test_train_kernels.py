@@ -167,3 +167,54 @@ def test_hip_ppo_learner_graph_trains():
         assert np.isfinite(v), (k, v)
     for p in pol.parameters():
         assert torch.isfinite(p).all()
+
+
+@pytest.mark.gpu
+def test_hip_adam_matches_torch_adam():
+    """Fused clip+Adam vs torch clip_grad_norm_ + Adam over 3 steps."""
+    from rlgpuschedule_amd.ops.train import HipAdam
+    torch.manual_seed(11)
+    D, H, A, HEADC = 84, 256, 17, 32
+    # params in torch layout
+    w1 = torch.randn(H, D, device="cuda")
+    b1 = torch.randn(H, device="cuda")
+    piw = torch.randn(A, H, device="cuda")
+    vw = torch.randn(1, H, device="cuda")
+    ref = {k: v.clone() for k, v in
+           dict(w1=w1, b1=b1, piw=piw, vw=vw).items()}
+    n = D * H + H + H * HEADC
+    gflat = torch.zeros(n, device="cuda")
+    dW1 = gflat[: D * H].view(D, H)
+    db1 = gflat[D * H: D * H + H]
+    dWh = gflat[D * H + H:].view(H, HEADC)
+    segs = [
+        dict(goff=0, R=D, C=H, c0=0, c1=H, param=w1, transposed=True),
+        dict(goff=D * H, R=1, C=H, c0=0, c1=H, param=b1, transposed=False),
+        dict(goff=D * H + H, R=H, C=HEADC, c0=0, c1=A, param=piw,
+             transposed=True),
+        dict(goff=D * H + H, R=H, C=HEADC, c0=A, c1=A + 1, param=vw,
+             transposed=True),
+    ]
+    opt = HipAdam(gflat, segs, lr=1e-2, eps=1e-5, max_norm=0.5)
+    refp = [torch.nn.Parameter(ref["w1"]), torch.nn.Parameter(ref["b1"]),
+            torch.nn.Parameter(ref["piw"]), torch.nn.Parameter(ref["vw"])]
+    topt = torch.optim.Adam(refp, lr=1e-2, eps=1e-5)
+    g = torch.Generator(device="cuda").manual_seed(3)
+    for step in range(3):
+        dW1.copy_(torch.randn(D, H, device="cuda", generator=g))
+        db1.copy_(torch.randn(H, device="cuda", generator=g))
+        dWh.zero_()
+        dWh[:, :A].copy_(torch.randn(H, A, device="cuda", generator=g))
+        dWh[:, A].copy_(torch.randn(H, device="cuda", generator=g))
+        # torch side: same grads in param layout
+        refp[0].grad = dW1.t().contiguous()
+        refp[1].grad = db1.clone()
+        refp[2].grad = dWh[:, :A].t().contiguous()
+        refp[3].grad = dWh[:, A:A + 1].t().contiguous()
+        opt.step()
+        torch.nn.utils.clip_grad_norm_(refp, 0.5)
+        topt.step()
+    torch.testing.assert_close(w1, refp[0].detach(), atol=1e-6, rtol=1e-5)
+    torch.testing.assert_close(b1, refp[1].detach(), atol=1e-6, rtol=1e-5)
+    torch.testing.assert_close(piw, refp[2].detach(), atol=1e-6, rtol=1e-5)
+    torch.testing.assert_close(vw, refp[3].detach(), atol=1e-6, rtol=1e-5)
