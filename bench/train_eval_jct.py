@@ -21,8 +21,12 @@ from rlgpuschedule_amd.trainer import CONFIGS, make_policy  # noqa: E402
 
 
 def run(iters=300, envs=2048, eval_envs=16, eval_eps=16, jobs=96,
-        interarrival=90.0, duration=600.0, seed=7):
-    """Programmatic version returning the full comparison dict."""
+        interarrival=90.0, duration=600.0, seed=7, phase2_iters=0,
+        phase2_ent=0.002, phase2_lr=1e-4):
+    """Programmatic version returning the full comparison dict.
+    Optional phase 2: entropy/lr anneal via a fresh learner (Adam state
+    resets; the captured graph bakes coefficients, so annealing means a
+    new capture)."""
     from rlgpuschedule_amd.agents.ppo import PPOConfig
     from rlgpuschedule_amd.agents.ppo_hip import HipPPOLearner
     from rlgpuschedule_amd.agents.rollout_gpu import GpuRolloutEngine
@@ -41,6 +45,13 @@ def run(iters=300, envs=2048, eval_envs=16, eval_eps=16, jobs=96,
     for it in range(iters):
         buf = engine.collect()
         stats = learner.update(buf)
+    if phase2_iters:
+        learner = HipPPOLearner(
+            policy, PPOConfig(epochs=1, minibatches=4, lr=phase2_lr,
+                              ent_coef=phase2_ent))
+        for it in range(phase2_iters):
+            buf = engine.collect()
+            stats = learner.update(buf)
     torch.cuda.synchronize()
     res = {"train_s": round(time.perf_counter() - t0, 1),
            "final": {k: round(v, 4) for k, v in stats.items()}}
