@@ -276,20 +276,14 @@ __global__ __launch_bounds__(256) void bwd_dw_kernel(BwdDwArgs p) {
   f32x4 acc[4] = {};
   const int wm = w * 16;             // wave's 16 m-rows within the tile
   const int ar = lane & 15, kk = lane >> 4, bc = lane & 15;
-  // T14 async-STAGE split (cdna_hip_programming.md G15): each thread owns
-  // 8 (k, j-quad) staging slots per 128-deep chunk; chunk c+1's GLOBAL
-  // loads are issued BEFORE chunk c's MFMAs (HBM latency hides under
-  // compute), the LDS writes land after the post-compute barrier. The
-  // transposed kperm LDS image forces register staging (global_load_lds
-  // writes lane-linear destinations only).
-  const int nchunks = (int)(p.BC / DW_KC);
-  f32x4 sx[8], sy[8];
-  auto issue_loads = [&](int kc) {
-    #pragma unroll
-    for (int q = 0; q < 8; ++q) {
-      int ix = tid + q * 256;
+  for (int kc = 0; kc < (int)p.BC; kc += DW_KC) {
+    // stage X^T and dY^T chunks with float4 global loads: thread ix covers
+    // (k, j4): 4 consecutive features j of batch row b -> 4 LDS rows at
+    // column kperm(k). 128 k x 16 j-quads = 2048 quads / 256 threads.
+    for (int ix = tid; ix < DW_KC * 16; ix += 256) {
       int k = ix >> 4, j4 = (ix & 15) * 4;
       int64_t b = b0 + kc + k;
+      int kp = kperm(k, DW_KC);
       f32x4 xv = {}, yv = {};
       if (b < p.B) {
         if (m0 + j4 + 3 < M) xv = *(const f32x4*)(p.X + b * M + m0 + j4);
@@ -301,28 +295,15 @@ __global__ __launch_bounds__(256) void bwd_dw_kernel(BwdDwArgs p) {
           for (int t = 0; t < 4; ++t)
             if (n0 + j4 + t < N) yv[t] = p.dY[b * N + n0 + j4 + t];
       }
-      sx[q] = xv;
-      sy[q] = yv;
-    }
-  };
-  auto write_lds = [&]() {
-    #pragma unroll
-    for (int q = 0; q < 8; ++q) {
-      int ix = tid + q * 256;
-      int k = ix >> 4, j4 = (ix & 15) * 4;
-      int kp = kperm(k, DW_KC);
       #pragma unroll
       for (int t = 0; t < 4; ++t) {
-        xt[(j4 + t) * DW_LD + kp] = sx[q][t];
-        yt[(j4 + t) * DW_LD + kp] = sy[q][t];
+        xt[(j4 + t) * DW_LD + kp] = xv[t];
+        yt[(j4 + t) * DW_LD + kp] = yv[t];
       }
     }
-  };
-  issue_loads(0);
-  write_lds();
-  __syncthreads();
-  for (int c = 0; c < nchunks; ++c) {
-    if (c + 1 < nchunks) issue_loads((c + 1) * DW_KC);
+    __syncthreads();
+    // C[wm..wm+15][0..63] over the 128-deep chunk; fragments are float4
+    // LDS reads (kperm: k = 16*k16 + 4s + kk -> kk*(KC/4) + 4*k16 + s)
     const float* arow = xt + (wm + ar) * DW_LD + kk * (DW_KC / 4);
     const float* brows = yt + bc * DW_LD + kk * (DW_KC / 4);
     #pragma unroll
@@ -338,10 +319,6 @@ __global__ __launch_bounds__(256) void bwd_dw_kernel(BwdDwArgs p) {
       }
     }
     __syncthreads();
-    if (c + 1 < nchunks) {
-      write_lds();
-      __syncthreads();
-    }
   }
   const int crow = (lane >> 4) * 4, ccol = lane & 15;
   #pragma unroll
