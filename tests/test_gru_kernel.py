@@ -101,3 +101,29 @@ def test_gru_bf16_actor_close_to_torch():
     assert (actions == a_ref).float().mean().item() > 0.95
     chosen = mask.gather(1, actions.unsqueeze(1))
     assert (chosen == 1).all()
+
+
+@pytest.mark.gpu
+def test_gru_sharded_engine():
+    import torch
+    from rlgpuschedule_amd.agents.ppo import PPOConfig
+    from rlgpuschedule_amd.agents.ppo_gru_hip import HipGRUPPOLearner
+    from rlgpuschedule_amd.agents.rollout_gpu import GpuRolloutEngine
+    from rlgpuschedule_amd.env.hip_env import HipVectorEnv
+    from rlgpuschedule_amd.env.spec import cluster64
+    from rlgpuschedule_amd.models.policy import GRUPolicy
+    spec = cluster64(horizon=50_000)
+    envs = [HipVectorEnv(spec, 128, trace_kind="philly", jobs_per_env=24,
+                         seed=21 + s) for s in range(2)]
+    policy = GRUPolicy(spec.obs_dim, spec.num_actions, 256).cuda()
+    eng = GpuRolloutEngine(envs, policy, T=16, seed=1, use_graph=True,
+                           actor_dtype="bf16")
+    learner = HipGRUPPOLearner(policy, PPOConfig(epochs=1, minibatches=2),
+                               use_graph=False)
+    for _ in range(3):
+        buf = eng.collect()
+        stats = learner.update(buf)
+    assert np.isfinite(stats["pi_loss"])
+    for t in range(16):
+        chosen = buf.mask[t].gather(1, buf.actions[t].unsqueeze(1))
+        assert (chosen == 1).all()
