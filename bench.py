@@ -46,6 +46,8 @@ def main():
     ap.add_argument("--hidden", type=int, default=256)
     ap.add_argument("--policy", choices=["mlp", "gru"], default="mlp")
     ap.add_argument("--actor-dtype", choices=["fp32", "bf16"], default="bf16")
+    ap.add_argument("--update-dtype", choices=["fp32", "bf16"],
+                    default="fp32")
     ap.add_argument("--minibatches", type=int, default=4)
     ap.add_argument("--shards", type=int, default=2,
                     help="env shards on separate HIP streams (overlaps "
@@ -89,10 +91,12 @@ def main():
         from rlgpuschedule_amd.agents.ppo_gru_hip import HipGRUPPOLearner
         from rlgpuschedule_amd.agents.rollout_gpu import GpuRolloutEngine
         if policy.is_recurrent:
-            lcls = HipGRUPPOLearner
+            learner = HipGRUPPOLearner(policy, cfg, grad_sync)
+        elif args.learner == "hip":
+            learner = HipPPOLearner(policy, cfg, grad_sync,
+                                    dtype=args.update_dtype)
         else:
-            lcls = HipPPOLearner if args.learner == "hip" else FusedPPOLearner
-        learner = lcls(policy, cfg, grad_sync)
+            learner = FusedPPOLearner(policy, cfg, grad_sync)
         collector = GpuRolloutEngine(env, policy, T, seed=555 + rank,
                                      actor_dtype=args.actor_dtype)
     else:
@@ -145,8 +149,10 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,  # no published reference numbers
-            "dtype": ("mixed (bf16 actor, fp32 update)"
-                      if args.actor_dtype == "bf16" else "fp32"),
+            "dtype": (f"mixed (actor {args.actor_dtype}, "
+                      f"update {args.update_dtype})"
+                      if "bf16" in (args.actor_dtype, args.update_dtype)
+                      else "fp32"),
             "data": "synthetic",
             "mean_jct_us": round(st.get("mean_jct_us", 0.0), 1),
             "episodes": st.get("episodes", 0),

@@ -19,10 +19,10 @@ from __future__ import annotations
 
 import torch
 
-from ..models.hip_train import HEADC, TrainMirror
+from ..models.hip_train import HEADC, TrainMirror, TrainMirrorBf16
 from ..ops.nn import gae_scan
-from ..ops.train import (bwd_dw, bwd_dx, colsum, fwd_train, gather_mb,
-                         ppo_head)
+from ..ops.train import (bwd_dw, bwd_dw_bf16, bwd_dx, bwd_dx_bf16, colsum,
+                         fwd_train, fwd_train_bf16, gather_mb, ppo_head)
 from .ppo import PPOConfig
 from .rollout import RolloutBuffer
 
@@ -69,7 +69,13 @@ class HipPPOLearner:
             "o": z(mb, D), "m": torch.zeros(mb, A, dtype=torch.uint8, device=dev),
             "a": torch.zeros(mb, dtype=torch.int64, device=dev),
             "olp": z(mb), "ov": z(mb), "advm": z(mb), "retm": z(mb),
-            "h1": z(mb, H), "h2": z(mb, H), "head": z(mb, HEADC),
+            "h1": torch.zeros(mb, H, device=dev,
+                              dtype=(torch.bfloat16 if self.dtype == "bf16"
+                                     else torch.float32)),
+            "h2": torch.zeros(mb, H, device=dev,
+                              dtype=(torch.bfloat16 if self.dtype == "bf16"
+                                     else torch.float32)),
+            "head": z(mb, HEADC),
             "dhead": z(mb, HEADC), "dh2": z(mb, H), "dh1": z(mb, H),
         }
         # grad buffers as views of ONE flat tensor -> a single zero_() per
@@ -156,17 +162,29 @@ class HipPPOLearner:
                 gather_mb(ix, obs, mask, acts, old_logp, old_v,
                           adv.contiguous(), ret, s["o"], s["m"], s["a"],
                           s["olp"], s["ov"], s["advm"], s["retm"])
-                fwd_train(s["o"], mi.wt1, mi.b1, mi.wt2, mi.b2, mi.wth,
-                          mi.bh, s["h1"], s["h2"], s["head"])
+                if self.dtype == "bf16":
+                    fwd_train_bf16(s["o"], mi.wt1, mi.b1, mi.wt2, mi.b2,
+                                   mi.wth, mi.bh, s["h1"], s["h2"], s["head"])
+                else:
+                    fwd_train(s["o"], mi.wt1, mi.b1, mi.wt2, mi.b2, mi.wth,
+                              mi.bh, s["h1"], s["h2"], s["head"])
                 ppo_head(s["head"], s["m"], s["a"], s["olp"], s["ov"],
                          s["advm"], s["retm"], cfg.clip, cfg.vf_coef,
                          cfg.ent_coef, s["dhead"], s["stats"])
-                bwd_dx(s["dhead"], mi.whT, s["h2"], s["dh2"])
-                bwd_dx(s["dh2"], mi.w2T, s["h1"], s["dh1"])
-                s["gflat"].zero_()
-                bwd_dw(s["h2"], s["dhead"], s["dWh"])
-                bwd_dw(s["h1"], s["dh2"], s["dW2"])
-                bwd_dw(s["o"], s["dh1"], s["dW1"])
+                if self.dtype == "bf16":
+                    bwd_dx_bf16(s["dhead"], mi.whT, s["h2"], s["dh2"])
+                    bwd_dx_bf16(s["dh2"], mi.w2T, s["h1"], s["dh1"])
+                    s["gflat"].zero_()
+                    bwd_dw_bf16(s["h2"], s["dhead"], s["dWh"])
+                    bwd_dw_bf16(s["h1"], s["dh2"], s["dW2"])
+                    bwd_dw_bf16(s["o"], s["dh1"], s["dW1"])
+                else:
+                    bwd_dx(s["dhead"], mi.whT, s["h2"], s["dh2"])
+                    bwd_dx(s["dh2"], mi.w2T, s["h1"], s["dh1"])
+                    s["gflat"].zero_()
+                    bwd_dw(s["h2"], s["dhead"], s["dWh"])
+                    bwd_dw(s["h1"], s["dh2"], s["dW2"])
+                    bwd_dw(s["o"], s["dh1"], s["dW1"])
                 colsum(s["dhead"], s["dbh"])
                 colsum(s["dh2"], s["db2"])
                 colsum(s["dh1"], s["db1"])

@@ -20,6 +20,37 @@ __device__ __forceinline__ int kperm(int k, int K) {
   return (k & 3) * (K >> 2) + (k >> 2);
 }
 
+// bf16 fragment scheme: both operands share the CONTIGUOUS per-lane k-map
+// k = kk*8 + j (any bijective shared map yields the correct dot product),
+// so a lane reads ONE bf16x8 (16 B) per 32-deep K step. Packed weights:
+// wt[K,N].view(K/32, 4(kk), 8(j), N).permute(0,1,3,2).contiguous().
+typedef __bf16 rls_bf16;
+typedef rls_bf16 bf16x8 __attribute__((ext_vector_type(8)));
+
+// C[16 x 64] for one wave on mfma_f32_16x16x32_bf16: A is a PLAIN
+// row-major bf16 LDS tile (contiguous k; lda in elements, keep
+// (lda/2) % 64 != 0 for bank spread), B packed global bf16.
+__device__ __forceinline__ void mfma_bf16_core_16x64(
+    const rls_bf16* __restrict__ a_lds, int lda, int K,
+    const rls_bf16* __restrict__ b_pk, int ldb, int col0, int lane,
+    f32x4 acc[4]) {
+  const int ar = lane & 15;
+  const int kk = lane >> 4;
+  const int bc = lane & 15;
+  const rls_bf16* arow = a_lds + ar * lda + kk * 8;
+  const rls_bf16* bbase = b_pk + (int64_t)8 * (kk * ldb + col0 + bc);
+  for (int k32 = 0; k32 < (K >> 5); ++k32) {
+    bf16x8 av = *(const bf16x8*)(arow + 32 * k32);
+    const rls_bf16* bk = bbase + (int64_t)32 * k32 * ldb;
+    #pragma unroll
+    for (int n = 0; n < 4; ++n) {
+      bf16x8 bv = *(const bf16x8*)(bk + 128 * n);
+      acc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(av, bv, acc[n],
+                                                       0, 0, 0);
+    }
+  }
+}
+
 // Core: C[16 x 64] += A[16 x K] * B[K x 64] for one wave.
 // A: LDS tile in kperm layout (row stride lda, floats). B: packed global
 // (logical row length ldb). acc: 4 n-subtiles of f32x4.

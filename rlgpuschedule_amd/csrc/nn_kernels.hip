@@ -671,8 +671,7 @@ extern "C" int rlsched_gru_actor(const int64_t* ip, const uint64_t* bufs,
 // .permute(0,1,3,2) -> lane (kk,bc) reads ONE bf16x8 per 32-deep K step.
 namespace {
 
-typedef __bf16 bf16;
-typedef bf16 bf16x8 __attribute__((ext_vector_type(8)));
+typedef rls_bf16 bf16;
 
 struct ActorBf16Args {
   int64_t E, D, H, A;

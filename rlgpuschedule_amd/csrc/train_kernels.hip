@@ -636,3 +636,286 @@ extern "C" int rlsched_adam_step(const int64_t* ip, const double* fp,
                      (hipStream_t)stream, a);
   return (int)hipGetLastError();
 }
+
+// --------------------------- bf16 mixed-precision update-path kernels
+// Opt-in (bench --update-dtype bf16): forward and backward GEMMs on
+// mfma_f32_16x16x32_bf16 (~13x the f32 rate), f32 accumulation, f32 loss
+// head and f32 weight gradients/Adam (activations h1/h2 stored bf16 —
+// half the save/reload traffic). Numerics: mixed-precision class; the
+// fp32 path stays the exactness default.
+namespace {
+
+struct FwdTrainBf16Args {
+  int64_t B, D, H;
+  const float* obs;
+  const rls_bf16* wt1;   // packed8 [Dp,H]
+  const float* b1;
+  const rls_bf16* wt2;   // packed8 [H,H]
+  const float* b2;
+  const rls_bf16* wth;   // packed8 [H,HEADC]
+  const float* bh;
+  rls_bf16* h1;          // [B,H] bf16 out
+  rls_bf16* h2;          // [B,H] bf16 out
+  float* head;           // [B,HEADC] f32 out
+};
+
+__global__ __launch_bounds__(256) void fwd_train_bf16_kernel(
+    FwdTrainBf16Args p) {
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int w = tid >> 6;
+  const int64_t row0 = (int64_t)blockIdx.x * ROWS;
+  const int D = (int)p.D, H = (int)p.H;
+  const int Dp = (D + 31) & ~31;
+  const int lda_o = Dp + 8, lda_h = H + 8;
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  rls_bf16* sobs = (rls_bf16*)smem;
+  rls_bf16* sh1 = sobs + ROWS * lda_o;
+  rls_bf16* sh2 = sh1 + ROWS * lda_h;
+  for (int ix = tid; ix < ROWS * Dp; ix += 256) {
+    int r = ix / Dp, d = ix % Dp;
+    float vv = 0.0f;
+    if (d < D && row0 + r < p.B) vv = p.obs[(row0 + r) * D + d];
+    sobs[r * lda_o + d] = (rls_bf16)vv;
+  }
+  __syncthreads();
+  for (int c0 = w * 64; c0 < H; c0 += 256) {
+    f32x4 acc[4] = {};
+    mfma_bf16_core_16x64(sobs, lda_o, Dp, p.wt1, H, c0, lane, acc);
+    const int crow = (lane >> 4) * 4, ccol = lane & 15;
+    #pragma unroll
+    for (int n = 0; n < 4; ++n)
+      #pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        int col = c0 + n * 16 + ccol;
+        rls_bf16 vv = (rls_bf16)tanhf(acc[n][i] + p.b1[col]);
+        sh1[(crow + i) * lda_h + col] = vv;
+        if (row0 + crow + i < p.B) p.h1[(row0 + crow + i) * H + col] = vv;
+      }
+  }
+  __syncthreads();
+  for (int c0 = w * 64; c0 < H; c0 += 256) {
+    f32x4 acc[4] = {};
+    mfma_bf16_core_16x64(sh1, lda_h, H, p.wt2, H, c0, lane, acc);
+    const int crow = (lane >> 4) * 4, ccol = lane & 15;
+    #pragma unroll
+    for (int n = 0; n < 4; ++n)
+      #pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        int col = c0 + n * 16 + ccol;
+        rls_bf16 vv = (rls_bf16)tanhf(acc[n][i] + p.b2[col]);
+        sh2[(crow + i) * lda_h + col] = vv;
+        if (row0 + crow + i < p.B) p.h2[(row0 + crow + i) * H + col] = vv;
+      }
+  }
+  __syncthreads();
+  if (w == 0) {
+    f32x4 acc[2] = {};
+    const int ar = lane & 15, kk = lane >> 4, bc = lane & 15;
+    const rls_bf16* arow = sh2 + ar * lda_h + kk * 8;
+    const rls_bf16* bbase = p.wth + (int64_t)8 * (kk * HEADC + bc);
+    for (int k32 = 0; k32 < (H >> 5); ++k32) {
+      bf16x8 av = *(const bf16x8*)(arow + 32 * k32);
+      const rls_bf16* bk = bbase + (int64_t)32 * k32 * HEADC;
+      #pragma unroll
+      for (int n = 0; n < 2; ++n) {
+        bf16x8 bv = *(const bf16x8*)(bk + 128 * n);
+        acc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(av, bv, acc[n],
+                                                         0, 0, 0);
+      }
+    }
+    const int crow = (lane >> 4) * 4, ccol = lane & 15;
+    #pragma unroll
+    for (int n = 0; n < 2; ++n)
+      #pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        int col = n * 16 + ccol;
+        if (row0 + crow + i < p.B)
+          p.head[(row0 + crow + i) * HEADC + col] = acc[n][i] + p.bh[col];
+      }
+  }
+}
+
+// dX = (dY @ W^T) [* (1 - h^2)]: dY f32 staged to bf16, W packed8 bf16,
+// h bf16 (null -> plain GEMM), dX f32.
+struct BwdDxBf16Args {
+  int64_t B, K, N;
+  const float* dY;
+  const rls_bf16* wtp;
+  const rls_bf16* h;
+  float* dX;
+};
+
+__global__ __launch_bounds__(256) void bwd_dx_bf16_kernel(BwdDxBf16Args p) {
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int w = tid >> 6;
+  const int64_t row0 = (int64_t)blockIdx.x * ROWS;
+  const int K = (int)p.K, N = (int)p.N;
+  const int lda = K + 8;
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  rls_bf16* sdy = (rls_bf16*)smem;    // [ROWS][K+8] plain
+  for (int ix = tid; ix < ROWS * K; ix += 256) {
+    int r = ix / K, k = ix % K;
+    float vv = 0.0f;
+    if (row0 + r < p.B) vv = p.dY[(row0 + r) * K + k];
+    sdy[r * lda + k] = (rls_bf16)vv;
+  }
+  __syncthreads();
+  for (int c0 = w * 64; c0 < N; c0 += 256) {
+    f32x4 acc[4] = {};
+    mfma_bf16_core_16x64(sdy, lda, K, p.wtp, N, c0, lane, acc);
+    const int crow = (lane >> 4) * 4, ccol = lane & 15;
+    #pragma unroll
+    for (int n = 0; n < 4; ++n)
+      #pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        int col = c0 + n * 16 + ccol;
+        int64_t row = row0 + crow + i;
+        if (row < p.B) {
+          float g = acc[n][i];
+          if (p.h) {
+            float hv = (float)p.h[row * N + col];
+            g *= (1.0f - hv * hv);
+          }
+          p.dX[row * N + col] = g;
+        }
+      }
+  }
+}
+
+// dW[M,N] += X^T @ dY over batch: operands staged TRANSPOSED into bf16
+// LDS tiles ([64 features][DW_KC k] plain contiguous-k layout), bf16 MFMA,
+// f32 atomic accumulation. X may be bf16 (saved activations) or f32 (obs);
+// dY f32 (upstream grads), converted during staging.
+struct BwdDwBf16Args {
+  int64_t B, M, N, BC;
+  const float* Xf;          // one of Xf/Xb is null
+  const rls_bf16* Xb;
+  const float* dY;
+  float* dW;
+};
+
+#define DWB_LD (DW_KC + 8)   // bf16 elements; /2 dwords % 64 == 4
+
+__global__ __launch_bounds__(256) void bwd_dw_bf16_kernel(BwdDwBf16Args p) {
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int w = tid >> 6;
+  const int M = (int)p.M, N = (int)p.N;
+  const int mtiles = (M + 63) >> 6, ntiles = (N + 63) >> 6;
+  int bid = blockIdx.x;
+  const int m0 = (bid % mtiles) * 64; bid /= mtiles;
+  const int n0 = (bid % ntiles) * 64; bid /= ntiles;
+  const int64_t b0 = (int64_t)bid * p.BC;
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  rls_bf16* xt = (rls_bf16*)smem;          // [64 m][DWB_LD]
+  rls_bf16* yt = xt + 64 * DWB_LD;         // [64 n][DWB_LD]
+  f32x4 acc[4] = {};
+  const int wm = w * 16;
+  const int ar = lane & 15, kk = lane >> 4, bc = lane & 15;
+  for (int kc = 0; kc < (int)p.BC; kc += DW_KC) {
+    for (int ix = tid; ix < DW_KC * 16; ix += 256) {
+      int k = ix >> 4, j4 = (ix & 15) * 4;
+      int64_t b = b0 + kc + k;
+      float xv[4] = {}, yv[4] = {};
+      if (b < p.B) {
+        for (int t = 0; t < 4; ++t) {
+          if (m0 + j4 + t < M)
+            xv[t] = p.Xf ? p.Xf[b * M + m0 + j4 + t]
+                         : (float)p.Xb[b * M + m0 + j4 + t];
+          if (n0 + j4 + t < N) yv[t] = p.dY[b * N + n0 + j4 + t];
+        }
+      }
+      #pragma unroll
+      for (int t = 0; t < 4; ++t) {
+        xt[(j4 + t) * DWB_LD + k] = (rls_bf16)xv[t];
+        yt[(j4 + t) * DWB_LD + k] = (rls_bf16)yv[t];
+      }
+    }
+    __syncthreads();
+    const rls_bf16* arow = xt + (wm + ar) * DWB_LD + kk * 8;
+    const rls_bf16* brows = yt + bc * DWB_LD + kk * 8;
+    #pragma unroll
+    for (int k32 = 0; k32 < DW_KC / 32; ++k32) {
+      bf16x8 av = *(const bf16x8*)(arow + 32 * k32);
+      #pragma unroll
+      for (int n = 0; n < 4; ++n) {
+        bf16x8 bv = *(const bf16x8*)(brows + n * 16 * DWB_LD + 32 * k32);
+        acc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(av, bv, acc[n],
+                                                         0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+  const int crow = (lane >> 4) * 4, ccol = lane & 15;
+  #pragma unroll
+  for (int n = 0; n < 4; ++n)
+    #pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      int m = m0 + wm + crow + i;
+      int nn = n0 + n * 16 + ccol;
+      if (m < M && nn < N) atomicAdd(&p.dW[(int64_t)m * N + nn], acc[n][i]);
+    }
+}
+
+}  // namespace
+
+extern "C" int rlsched_fwd_train_bf16(const int64_t* ip, const uint64_t* bufs,
+                                      void* stream) {
+  FwdTrainBf16Args p;
+  p.B = ip[0]; p.D = ip[1]; p.H = ip[2];
+  if (p.H % 64) return -1;
+  p.obs = (const float*)bufs[0];
+  p.wt1 = (const rls_bf16*)bufs[1];
+  p.b1 = (const float*)bufs[2];
+  p.wt2 = (const rls_bf16*)bufs[3];
+  p.b2 = (const float*)bufs[4];
+  p.wth = (const rls_bf16*)bufs[5];
+  p.bh = (const float*)bufs[6];
+  p.h1 = (rls_bf16*)bufs[7];
+  p.h2 = (rls_bf16*)bufs[8];
+  p.head = (float*)bufs[9];
+  const int Dp = ((int)p.D + 31) & ~31;
+  size_t lds = sizeof(rls_bf16)
+             * (ROWS * (Dp + 8) + 2 * ROWS * ((int)p.H + 8));
+  dim3 grid((unsigned)((p.B + ROWS - 1) / ROWS));
+  hipLaunchKernelGGL(fwd_train_bf16_kernel, grid, dim3(256), lds,
+                     (hipStream_t)stream, p);
+  return (int)hipGetLastError();
+}
+
+extern "C" int rlsched_bwd_dx_bf16(const int64_t* ip, const uint64_t* bufs,
+                                   void* stream) {
+  BwdDxBf16Args p;
+  p.B = ip[0]; p.K = ip[1]; p.N = ip[2];
+  if (p.K % 32 || p.N % 64) return -1;
+  p.dY = (const float*)bufs[0];
+  p.wtp = (const rls_bf16*)bufs[1];
+  p.h = (const rls_bf16*)bufs[2];
+  p.dX = (float*)bufs[3];
+  size_t lds = sizeof(rls_bf16) * ROWS * ((int)p.K + 8);
+  dim3 grid((unsigned)((p.B + ROWS - 1) / ROWS));
+  hipLaunchKernelGGL(bwd_dx_bf16_kernel, grid, dim3(256), lds,
+                     (hipStream_t)stream, p);
+  return (int)hipGetLastError();
+}
+
+extern "C" int rlsched_bwd_dw_bf16(const int64_t* ip, const uint64_t* bufs,
+                                   void* stream) {
+  BwdDwBf16Args p;
+  p.B = ip[0]; p.M = ip[1]; p.N = ip[2]; p.BC = ip[3];
+  const int64_t x_is_bf16 = ip[4];
+  if (p.BC % DW_KC) return -1;
+  p.Xf = x_is_bf16 ? nullptr : (const float*)bufs[0];
+  p.Xb = x_is_bf16 ? (const rls_bf16*)bufs[0] : nullptr;
+  p.dY = (const float*)bufs[1];
+  p.dW = (float*)bufs[2];
+  const int mtiles = ((int)p.M + 63) >> 6, ntiles = ((int)p.N + 63) >> 6;
+  const int chunks = (int)((p.B + p.BC - 1) / p.BC);
+  size_t lds = sizeof(rls_bf16) * 2 * 64 * DWB_LD;
+  dim3 grid((unsigned)(mtiles * ntiles * chunks));
+  hipLaunchKernelGGL(bwd_dw_bf16_kernel, grid, dim3(256), lds,
+                     (hipStream_t)stream, p);
+  return (int)hipGetLastError();
+}

@@ -193,3 +193,52 @@ class HipAdam:
         self.m.copy_(sd["m"].to(self.m.device))
         self.v.copy_(sd["v"].to(self.v.device))
         self.t.copy_(sd["t"].to(self.t.device))
+
+
+def _load_bf16():
+    lib = load()
+    if not hasattr(lib, "_tbf16_ready"):
+        for fn in ("rlsched_fwd_train_bf16", "rlsched_bwd_dx_bf16",
+                   "rlsched_bwd_dw_bf16"):
+            f = getattr(lib, fn)
+            f.restype = ctypes.c_int
+            f.argtypes = [ctypes.c_void_p] * 3
+        lib._tbf16_ready = True
+    return lib
+
+
+def fwd_train_bf16(obs, wt1, b1, wt2, b2, wth, bh, h1, h2, head):
+    """bf16-MFMA training forward; h1/h2 saved bf16, head f32."""
+    _load_bf16()
+    B, D = obs.shape
+    H = h1.shape[1]
+    _call("rlsched_fwd_train_bf16", [B, D, H],
+          _ptrs(obs, wt1, b1, wt2, b2, wth, bh, h1, h2, head), _stream(obs))
+
+
+def bwd_dx_bf16(dY, wtp, h, dX):
+    """dX = (dY @ W^T) [* (1-h^2)], bf16 MFMA. dY/dX f32, h bf16 or None."""
+    _load_bf16()
+    B, K = dY.shape
+    N = dX.shape[1]
+    hp = h.data_ptr() if h is not None else 0
+    ipa = np.asarray([B, K, N], np.int64)
+    bufs = np.array([dY.data_ptr(), wtp.data_ptr(), hp, dX.data_ptr()],
+                    dtype=np.uint64)
+    rc = load().rlsched_bwd_dx_bf16(
+        ipa.ctypes.data_as(ctypes.c_void_p),
+        bufs.ctypes.data_as(ctypes.c_void_p),
+        ctypes.c_void_p(_stream(dY)))
+    if rc != 0:
+        raise RuntimeError(f"rlsched_bwd_dx_bf16 failed: {rc}")
+
+
+def bwd_dw_bf16(X, dY, dW, bc: int = 1024):
+    """dW[M,N] = X^T @ dY, bf16 MFMA (X f32 or bf16; dY f32; dW f32,
+    caller zeroes)."""
+    _load_bf16()
+    B, M = X.shape
+    N = dY.shape[1]
+    _call("rlsched_bwd_dw_bf16",
+          [B, M, N, bc, 1 if X.dtype == torch.bfloat16 else 0],
+          _ptrs(X, dY, dW), _stream(dY))

@@ -218,3 +218,40 @@ def test_hip_adam_matches_torch_adam():
     torch.testing.assert_close(b1, refp[1].detach(), atol=1e-6, rtol=1e-5)
     torch.testing.assert_close(piw, refp[2].detach(), atol=1e-6, rtol=1e-5)
     torch.testing.assert_close(vw, refp[3].detach(), atol=1e-6, rtol=1e-5)
+
+
+@pytest.mark.gpu
+def test_bf16_update_path_close_to_fp32():
+    """Mixed-precision update: gradient DIRECTION must track the fp32 path
+    (cosine > 0.995 per weight tensor after one full-batch update), and a
+    few updates stay finite."""
+    from rlgpuschedule_amd.agents.ppo import PPOConfig
+    from rlgpuschedule_amd.agents.ppo_hip import HipPPOLearner
+    from rlgpuschedule_amd.models.policy import MLPPolicy
+    from helpers import make_buf
+    torch.manual_seed(5)
+    spec = cluster64()
+    T, E = 16, 128
+    pol_a = MLPPolicy(spec.obs_dim, spec.num_actions, 256).cuda()
+    pol_b = copy.deepcopy(pol_a)
+    p0 = {n: p.detach().clone() for n, p in pol_a.named_parameters()}
+    cfg = PPOConfig(epochs=1, minibatches=1, norm_reward=False)
+    l32 = HipPPOLearner(pol_a, cfg, use_graph=False, dtype="fp32")
+    l16 = HipPPOLearner(pol_b, cfg, use_graph=False, dtype="bf16")
+    buf = make_buf(T, E, spec, "cuda", seed=8)
+    l32.update(buf)
+    l16.update(buf)
+    for (n, pa), (_, pb) in zip(pol_a.named_parameters(),
+                                pol_b.named_parameters()):
+        da = (pa - p0[n]).flatten()
+        db = (pb - p0[n]).flatten()
+        cos = torch.nn.functional.cosine_similarity(da, db, dim=0).item()
+        assert cos > 0.995, (n, cos)
+    # multi-update stability (graph mode)
+    pol_c = MLPPolicy(spec.obs_dim, spec.num_actions, 256).cuda()
+    lg = HipPPOLearner(pol_c, PPOConfig(epochs=1, minibatches=2),
+                       use_graph=True, dtype="bf16")
+    for _ in range(3):
+        stats = lg.update(buf)
+    for k, v in stats.items():
+        assert np.isfinite(v), (k, v)
