@@ -29,10 +29,16 @@ from .rollout import RolloutBuffer
 
 class HipPPOLearner:
     def __init__(self, policy, cfg: PPOConfig, grad_sync=None,
-                 use_graph: bool = True, use_hip_adam: bool = True):
+                 use_graph: bool = True, use_hip_adam: bool = True,
+                 dtype: str = "fp32"):
+        """dtype="bf16": mixed-precision update (bf16 MFMA GEMMs +
+        bf16-saved activations, f32 loss/grads/Adam). "fp32" matches the
+        torch reference to gradient-check tolerance."""
         assert not policy.is_recurrent
+        assert dtype in ("fp32", "bf16")
         self.policy = policy
         self.cfg = cfg
+        self.dtype = dtype
         self.grad_sync = grad_sync
         self.use_hip_adam = use_hip_adam
         self.hip_opt = None
@@ -42,7 +48,8 @@ class HipPPOLearner:
         self.use_graph = use_graph
         self.opt = torch.optim.Adam(policy.parameters(), lr=cfg.lr, eps=1e-5,
                                     capturable=self.use_graph)
-        self.mirror = TrainMirror(policy)
+        self.mirror = (TrainMirrorBf16(policy) if dtype == "bf16"
+                       else TrainMirror(policy))
         self.gen = torch.Generator(device="cuda")
         self.gen.manual_seed(7654321)
         self._graph = None
