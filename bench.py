@@ -41,14 +41,14 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=12)
     ap.add_argument("--warmup", type=int, default=3)
-    ap.add_argument("--envs", type=int, default=8192, help="envs per GPU")
+    ap.add_argument("--envs", type=int, default=16384, help="envs per GPU")
     ap.add_argument("--rollout", type=int, default=64)
     ap.add_argument("--hidden", type=int, default=256)
     ap.add_argument("--policy", choices=["mlp", "gru"], default="mlp")
     ap.add_argument("--actor-dtype", choices=["fp32", "bf16"], default="bf16")
     ap.add_argument("--update-dtype", choices=["fp32", "bf16"],
-                    default="fp32")
-    ap.add_argument("--minibatches", type=int, default=4)
+                    default="bf16")
+    ap.add_argument("--minibatches", type=int, default=2)
     ap.add_argument("--shards", type=int, default=2,
                     help="env shards on separate HIP streams (overlaps "
                          "actor and env kernels across shards)")
