@@ -91,7 +91,8 @@ def main():
         from rlgpuschedule_amd.agents.ppo_gru_hip import HipGRUPPOLearner
         from rlgpuschedule_amd.agents.rollout_gpu import GpuRolloutEngine
         if policy.is_recurrent:
-            learner = HipGRUPPOLearner(policy, cfg, grad_sync)
+            learner = HipGRUPPOLearner(policy, cfg, grad_sync,
+                                       dtype=args.update_dtype)
         elif args.learner == "hip":
             learner = HipPPOLearner(policy, cfg, grad_sync,
                                     dtype=args.update_dtype)

@@ -20,27 +20,34 @@ from __future__ import annotations
 
 import torch
 
-from ..models.gru_train_mirror import HEADC, GruTrainMirror
-from ..ops.gru import gru_seq_fwd, gru_seq_bwd, head_fwd
+from ..models.gru_train_mirror import (HEADC, GruTrainMirror,
+                                       GruTrainMirrorBf16)
+from ..ops.gru import (gru_seq_fwd, gru_seq_bwd, gru_seq_fwd_bf16,
+                       gru_seq_bwd_bf16, head_fwd, head_fwd_bf16)
 from ..ops.nn import gae_scan
-from ..ops.train import bwd_dw, bwd_dx, colsum, ppo_head
+from ..ops.train import (bwd_dw, bwd_dw_bf16, bwd_dx, bwd_dx_bf16, colsum,
+                         ppo_head)
 from .ppo import PPOConfig
 from .rollout import RolloutBuffer
 
 
 class HipGRUPPOLearner:
     def __init__(self, policy, cfg: PPOConfig, grad_sync=None,
-                 use_graph: bool = True, use_hip_adam: bool = True):
+                 use_graph: bool = True, use_hip_adam: bool = True,
+                 dtype: str = "fp32"):
         assert policy.is_recurrent
+        assert dtype in ("fp32", "bf16")
         self.policy = policy
         self.cfg = cfg
+        self.dtype = dtype
         self.grad_sync = grad_sync
         self.use_hip_adam = use_hip_adam
         self.hip_opt = None
         self.use_graph = use_graph
         self.opt = torch.optim.Adam(policy.parameters(), lr=cfg.lr, eps=1e-5,
                                     capturable=self.use_graph)
-        self.mirror = GruTrainMirror(policy)
+        self.mirror = (GruTrainMirrorBf16(policy) if dtype == "bf16"
+                       else GruTrainMirror(policy))
         self.gen = torch.Generator(device="cuda")
         self.gen.manual_seed(24680)
         self._graph = None
@@ -72,8 +79,10 @@ class HipGRUPPOLearner:
             "olp": z(T, Emb), "ov": z(T, Emb),
             "advm": z(T, Emb), "retm": z(T, Emb),
             "h0": z(Emb, H),
-            "xs": z(B, H), "rs": z(B, H), "zs": z(B, H), "ns": z(B, H),
-            "ghns": z(B, H), "hprevs": z(B, H), "hs": z(B, H),
+            **{k: torch.zeros(B, H, device=dev,
+                              dtype=(torch.bfloat16 if self.dtype == "bf16"
+                                     else torch.float32))
+               for k in ("xs", "rs", "zs", "ns", "ghns", "hprevs", "hs")},
             "head": z(B, HEADC), "dhead": z(B, HEADC),
             "dh_ext": z(T, Emb, H),
             "dgi": z(B, 3 * H), "dgh": z(B, 3 * H),
@@ -164,29 +173,56 @@ class HipGRUPPOLearner:
                 torch.index_select(adv, 1, ix, out=s["advm"])
                 torch.index_select(s["ret"], 1, ix, out=s["retm"])
                 torch.index_select(buf.h0, 0, ix, out=s["h0"])
-                gru_seq_fwd(s["obs"], s["done"], s["h0"], mi.wtin, mi.bin,
-                            mi.wih, mi.bih, mi.whh, mi.bhh, s["xs"], s["rs"],
-                            s["zs"], s["ns"], s["ghns"], s["hprevs"], s["hs"])
-                head_fwd(s["hs"], mi.wth, mi.bh, s["head"])
+                if self.dtype == "bf16":
+                    gru_seq_fwd_bf16(s["obs"], s["done"], s["h0"], mi.wtin,
+                                     mi.bin, mi.wih, mi.bih, mi.whh, mi.bhh,
+                                     s["xs"], s["rs"], s["zs"], s["ns"],
+                                     s["ghns"], s["hprevs"], s["hs"])
+                    head_fwd_bf16(s["hs"], mi.wth, mi.bh, s["head"])
+                else:
+                    gru_seq_fwd(s["obs"], s["done"], s["h0"], mi.wtin,
+                                mi.bin, mi.wih, mi.bih, mi.whh, mi.bhh,
+                                s["xs"], s["rs"], s["zs"], s["ns"],
+                                s["ghns"], s["hprevs"], s["hs"])
+                    head_fwd(s["hs"], mi.wth, mi.bh, s["head"])
                 ppo_head(s["head"], s["mask"].view(B, A),
                          s["acts"].view(B), s["olp"].view(B),
                          s["ov"].view(B), s["advm"].reshape(B),
                          s["retm"].view(B), cfg.clip, cfg.vf_coef,
                          cfg.ent_coef, s["dhead"], s["stats"])
-                bwd_dx(s["dhead"], mi.whT, None, s["dh_ext"].view(B, H))
-                gru_seq_bwd(s["dh_ext"], s["done"], s["rs"], s["zs"],
-                            s["ns"], s["ghns"], s["hprevs"], mi.whh_bwd,
-                            s["dgi"], s["dgh"])
-                bwd_dx(s["dgi"], mi.wih_bwd, s["xs"], s["dxpre"])
-                s["gflat"].zero_()
-                bwd_dw(s["hs"], s["dhead"], s["dWh"])
-                colsum(s["dhead"], s["dbh"])
-                bwd_dw(s["xs"], s["dgi"], s["dWih"])
-                colsum(s["dgi"], s["dbih"])
-                bwd_dw(s["hprevs"], s["dgh"], s["dWhh"])
-                colsum(s["dgh"], s["dbhh"])
-                bwd_dw(s["obs"].view(B, self._D), s["dxpre"], s["dWin"])
-                colsum(s["dxpre"], s["dbin"])
+                if self.dtype == "bf16":
+                    bwd_dx_bf16(s["dhead"], mi.whT, None,
+                                s["dh_ext"].view(B, H))
+                    gru_seq_bwd_bf16(s["dh_ext"], s["done"], s["rs"],
+                                     s["zs"], s["ns"], s["ghns"],
+                                     s["hprevs"], mi.whh_bwd, s["dgi"],
+                                     s["dgh"])
+                    bwd_dx_bf16(s["dgi"], mi.wih_bwd, s["xs"], s["dxpre"])
+                    s["gflat"].zero_()
+                    bwd_dw_bf16(s["hs"], s["dhead"], s["dWh"])
+                    colsum(s["dhead"], s["dbh"])
+                    bwd_dw_bf16(s["xs"], s["dgi"], s["dWih"])
+                    colsum(s["dgi"], s["dbih"])
+                    bwd_dw_bf16(s["hprevs"], s["dgh"], s["dWhh"])
+                    colsum(s["dgh"], s["dbhh"])
+                    bwd_dw_bf16(s["obs"].view(B, self._D), s["dxpre"],
+                                s["dWin"])
+                    colsum(s["dxpre"], s["dbin"])
+                else:
+                    bwd_dx(s["dhead"], mi.whT, None, s["dh_ext"].view(B, H))
+                    gru_seq_bwd(s["dh_ext"], s["done"], s["rs"], s["zs"],
+                                s["ns"], s["ghns"], s["hprevs"], mi.whh_bwd,
+                                s["dgi"], s["dgh"])
+                    bwd_dx(s["dgi"], mi.wih_bwd, s["xs"], s["dxpre"])
+                    s["gflat"].zero_()
+                    bwd_dw(s["hs"], s["dhead"], s["dWh"])
+                    colsum(s["dhead"], s["dbh"])
+                    bwd_dw(s["xs"], s["dgi"], s["dWih"])
+                    colsum(s["dgi"], s["dbih"])
+                    bwd_dw(s["hprevs"], s["dgh"], s["dWhh"])
+                    colsum(s["dgh"], s["dbhh"])
+                    bwd_dw(s["obs"].view(B, self._D), s["dxpre"], s["dWin"])
+                    colsum(s["dxpre"], s["dbin"])
                 if self.hip_opt is not None:
                     self.hip_opt.step()   # fused clip+Adam on gflat
                 else:

@@ -369,3 +369,328 @@ extern "C" int rlsched_head_fwd(const int64_t* ip, const uint64_t* bufs,
                      (hipStream_t)stream, p);
   return (int)hipGetLastError();
 }
+
+// ------------------------- bf16 mixed-precision GRU BPTT (opt-in)
+// Same structure as the f32 pair; GEMMs on mfma_f32_16x16x32_bf16 with
+// bf16 LDS tiles (contiguous shared k-map), f32 gate math, an f32 LDS
+// carry for the exact z*h path, bf16-saved activations (xs/rs/zs/ns/
+// ghns/hprevs/hs), f32 gate-grad outputs (consumed by the bf16 dW/dx
+// batched kernels which convert during staging).
+namespace {
+
+struct GruFwdBf16Args {
+  int64_t T, Emb, D, H;
+  const float* obs;
+  const uint8_t* done;
+  const float* h0;
+  const rls_bf16* wtin;   // packed8 [Dp,H]
+  const float* bin;
+  const rls_bf16* wih;    // packed8 [H,3H]
+  const float* bih;
+  const rls_bf16* whh;    // packed8 [H,3H]
+  const float* bhh;
+  rls_bf16 *xs, *rs, *zs, *ns, *ghns, *hprevs, *hs;   // [T,Emb,H] bf16
+};
+
+__global__ __launch_bounds__(256) void gru_seq_fwd_bf16_kernel(
+    GruFwdBf16Args p) {
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int w = tid >> 6;
+  const int64_t r0 = (int64_t)blockIdx.x * ROWS;
+  const int D = (int)p.D, H = (int)p.H;
+  const int Dp = (D + 31) & ~31;
+  const int lda_o = Dp + 8, lda_h = H + 8;
+  const int64_t Emb = p.Emb;
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  rls_bf16* sobs = (rls_bf16*)smem;              // [ROWS][Dp+8]
+  rls_bf16* sx = sobs + ROWS * lda_o;            // [ROWS][H+8]
+  rls_bf16* shx = sx + ROWS * lda_h;             // [ROWS][H+8]
+  rls_bf16* snew = shx + ROWS * lda_h;           // [ROWS][H+8]
+  float* shxf = (float*)(snew + ROWS * lda_h);   // [ROWS][H] f32 carry
+  float* snewf = shxf + ROWS * H;                // [ROWS][H]
+
+  for (int ix = tid; ix < ROWS * H; ix += 256) {
+    int r = ix / H, d = ix % H;
+    float vv = 0.0f;
+    if (r0 + r < Emb) {
+      vv = p.h0[(r0 + r) * H + d];
+      p.hprevs[(r0 + r) * H + d] = (rls_bf16)vv;
+    }
+    shx[r * lda_h + d] = (rls_bf16)vv;
+    shxf[r * H + d] = vv;
+  }
+  __syncthreads();
+
+  for (int64_t t = 0; t < p.T; ++t) {
+    const int64_t base = t * Emb + r0;
+    for (int ix = tid; ix < ROWS * Dp; ix += 256) {
+      int r = ix / Dp, d = ix % Dp;
+      float vv = 0.0f;
+      if (d < D && r0 + r < Emb) vv = p.obs[(base + r) * D + d];
+      sobs[r * lda_o + d] = (rls_bf16)vv;
+    }
+    __syncthreads();
+    for (int c0 = w * 64; c0 < H; c0 += 256) {
+      f32x4 acc[4] = {};
+      mfma_bf16_core_16x64(sobs, lda_o, Dp, p.wtin, H, c0, lane, acc);
+      const int crow = (lane >> 4) * 4, ccol = lane & 15;
+      #pragma unroll
+      for (int n = 0; n < 4; ++n)
+        #pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          int col = c0 + n * 16 + ccol;
+          rls_bf16 vv = (rls_bf16)tanhf(acc[n][i] + p.bin[col]);
+          sx[(crow + i) * lda_h + col] = vv;
+          if (r0 + crow + i < Emb) p.xs[(base + crow + i) * H + col] = vv;
+        }
+    }
+    __syncthreads();
+    for (int c0 = w * 64; c0 < H; c0 += 256) {
+      f32x4 gir[4] = {}, giz[4] = {}, gin[4] = {};
+      f32x4 ghr[4] = {}, ghz[4] = {}, ghn[4] = {};
+      mfma_bf16_core_16x64(sx, lda_h, H, p.wih, 3 * H, 0 * H + c0, lane, gir);
+      mfma_bf16_core_16x64(sx, lda_h, H, p.wih, 3 * H, 1 * H + c0, lane, giz);
+      mfma_bf16_core_16x64(sx, lda_h, H, p.wih, 3 * H, 2 * H + c0, lane, gin);
+      mfma_bf16_core_16x64(shx, lda_h, H, p.whh, 3 * H, 0 * H + c0, lane, ghr);
+      mfma_bf16_core_16x64(shx, lda_h, H, p.whh, 3 * H, 1 * H + c0, lane, ghz);
+      mfma_bf16_core_16x64(shx, lda_h, H, p.whh, 3 * H, 2 * H + c0, lane, ghn);
+      const int crow = (lane >> 4) * 4, ccol = lane & 15;
+      #pragma unroll
+      for (int n = 0; n < 4; ++n)
+        #pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          int col = c0 + n * 16 + ccol;
+          int r = crow + i;
+          int64_t grow = base + r;
+          float rr = sigf(gir[n][i] + p.bih[col] + ghr[n][i] + p.bhh[col]);
+          float zz = sigf(giz[n][i] + p.bih[H + col]
+                          + ghz[n][i] + p.bhh[H + col]);
+          float gn = ghn[n][i] + p.bhh[2 * H + col];
+          float nn = tanhf(gin[n][i] + p.bih[2 * H + col] + rr * gn);
+          float hold = shxf[r * H + col];
+          float hnew = (1.0f - zz) * nn + zz * hold;
+          if (r0 + r < Emb) {
+            p.rs[grow * H + col] = (rls_bf16)rr;
+            p.zs[grow * H + col] = (rls_bf16)zz;
+            p.ns[grow * H + col] = (rls_bf16)nn;
+            p.ghns[grow * H + col] = (rls_bf16)gn;
+            p.hs[grow * H + col] = (rls_bf16)hnew;
+          }
+          float hmask = hnew;
+          if (r0 + r < Emb && p.done[t * Emb + r0 + r]) hmask = 0.0f;
+          snew[r * lda_h + col] = (rls_bf16)hmask;
+          snewf[r * H + col] = hmask;
+          if (t + 1 < p.T && r0 + r < Emb)
+            p.hprevs[((t + 1) * Emb + r0 + r) * H + col] = (rls_bf16)hmask;
+        }
+    }
+    __syncthreads();
+    for (int ix = tid; ix < ROWS * lda_h; ix += 256)
+      shx[ix] = snew[ix];
+    for (int ix = tid; ix < ROWS * H; ix += 256)
+      shxf[ix] = snewf[ix];
+    __syncthreads();
+  }
+}
+
+struct GruBwdBf16Args {
+  int64_t T, Emb, H;
+  const float* dh_ext;        // [T,Emb,H] f32
+  const uint8_t* done;
+  const rls_bf16 *rs, *zs, *ns, *ghns, *hprevs;   // bf16 saves
+  const rls_bf16* whh_bwd;    // packed8 [3H,H]
+  float* dgi_all;             // [T,Emb,3H] f32
+  float* dgh_all;             // [T,Emb,3H] f32
+};
+
+__global__ __launch_bounds__(256) void gru_seq_bwd_bf16_kernel(
+    GruBwdBf16Args p) {
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int w = tid >> 6;
+  const int64_t r0 = (int64_t)blockIdx.x * ROWS;
+  const int H = (int)p.H;
+  const int H3 = 3 * H;
+  const int lda_h = H + 4, lda_3 = H3 + 8;   // bf16 tile stride for sdgh
+  const int64_t Emb = p.Emb;
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  rls_bf16* sdgh = (rls_bf16*)smem;               // [ROWS][3H+8] bf16
+  float* scarry = (float*)(sdgh + ROWS * lda_3);  // [ROWS][H+4]
+  float* stmp = scarry + ROWS * lda_h;            // [ROWS][H+4]
+
+  for (int ix = tid; ix < ROWS * lda_h; ix += 256) {
+    scarry[ix] = 0.0f;
+    stmp[ix] = 0.0f;
+  }
+  __syncthreads();
+
+  for (int64_t t = p.T - 1; t >= 0; --t) {
+    const int64_t base = t * Emb + r0;
+    for (int ix = tid; ix < ROWS * H; ix += 256) {
+      int r = ix / H, col = ix % H;
+      int64_t grow = base + r;
+      float dh = scarry[r * lda_h + col];
+      float dgr = 0.f, dgz = 0.f, dgn = 0.f, dghn = 0.f, dhz = 0.f;
+      if (r0 + r < Emb) {
+        dh += p.dh_ext[grow * H + col];
+        float rr = (float)p.rs[grow * H + col];
+        float zz = (float)p.zs[grow * H + col];
+        float nn = (float)p.ns[grow * H + col];
+        float gn = (float)p.ghns[grow * H + col];
+        float hp = (float)p.hprevs[grow * H + col];
+        float dz = dh * (hp - nn);
+        float dn = dh * (1.0f - zz);
+        dhz = dh * zz;
+        float dn_pre = dn * (1.0f - nn * nn);
+        float dr = dn_pre * gn;
+        float dr_pre = dr * rr * (1.0f - rr);
+        float dz_pre = dz * zz * (1.0f - zz);
+        dgr = dr_pre;
+        dgz = dz_pre;
+        dgn = dn_pre;
+        dghn = dn_pre * rr;
+        p.dgi_all[grow * H3 + col] = dgr;
+        p.dgi_all[grow * H3 + H + col] = dgz;
+        p.dgi_all[grow * H3 + 2 * H + col] = dgn;
+        p.dgh_all[grow * H3 + col] = dgr;
+        p.dgh_all[grow * H3 + H + col] = dgz;
+        p.dgh_all[grow * H3 + 2 * H + col] = dghn;
+      }
+      stmp[r * lda_h + col] = dhz;
+      sdgh[r * lda_3 + col] = (rls_bf16)dgr;
+      sdgh[r * lda_3 + H + col] = (rls_bf16)dgz;
+      sdgh[r * lda_3 + 2 * H + col] = (rls_bf16)dghn;
+    }
+    __syncthreads();
+    for (int c0 = w * 64; c0 < H; c0 += 256) {
+      f32x4 acc[4] = {};
+      mfma_bf16_core_16x64(sdgh, lda_3, H3, p.whh_bwd, H, c0, lane, acc);
+      const int crow = (lane >> 4) * 4, ccol = lane & 15;
+      #pragma unroll
+      for (int n = 0; n < 4; ++n)
+        #pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          int col = c0 + n * 16 + ccol;
+          int r = crow + i;
+          float dprev = acc[n][i] + stmp[r * lda_h + col];
+          if (t > 0 && r0 + r < Emb
+              && p.done[(t - 1) * Emb + r0 + r]) dprev = 0.0f;
+          scarry[r * lda_h + col] = dprev;
+        }
+    }
+    __syncthreads();
+  }
+}
+
+// head[B,32] = h(bf16)[B,H] @ wth(packed8) + bh
+struct HeadFwdBf16Args {
+  int64_t B, H;
+  const rls_bf16* h;
+  const rls_bf16* wth;
+  const float* bh;
+  float* head;
+};
+
+__global__ __launch_bounds__(256) void head_fwd_bf16_kernel(
+    HeadFwdBf16Args p) {
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int w = tid >> 6;
+  const int64_t r0 = (int64_t)blockIdx.x * ROWS;
+  const int H = (int)p.H;
+  const int lda_h = H + 8;
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  rls_bf16* sh = (rls_bf16*)smem;
+  for (int ix = tid; ix < ROWS * H; ix += 256) {
+    int r = ix / H, d = ix % H;
+    sh[r * lda_h + d] = (r0 + r < p.B) ? p.h[(r0 + r) * H + d]
+                                       : (rls_bf16)0.0f;
+  }
+  __syncthreads();
+  if (w == 0) {
+    f32x4 acc[2] = {};
+    const int ar = lane & 15, kk = lane >> 4, bc = lane & 15;
+    const rls_bf16* arow = sh + ar * lda_h + kk * 8;
+    const rls_bf16* bbase = p.wth + (int64_t)8 * (kk * HEADC + bc);
+    for (int k32 = 0; k32 < (H >> 5); ++k32) {
+      bf16x8 av = *(const bf16x8*)(arow + 32 * k32);
+      const rls_bf16* bk = bbase + (int64_t)32 * k32 * HEADC;
+      #pragma unroll
+      for (int n = 0; n < 2; ++n) {
+        bf16x8 bv = *(const bf16x8*)(bk + 128 * n);
+        acc[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(av, bv, acc[n],
+                                                         0, 0, 0);
+      }
+    }
+    const int crow = (lane >> 4) * 4, ccol = lane & 15;
+    #pragma unroll
+    for (int n = 0; n < 2; ++n)
+      #pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        int col = n * 16 + ccol;
+        if (r0 + crow + i < p.B)
+          p.head[(r0 + crow + i) * HEADC + col] = acc[n][i] + p.bh[col];
+      }
+  }
+}
+
+}  // namespace
+
+extern "C" int rlsched_gru_seq_fwd_bf16(const int64_t* ip,
+                                        const uint64_t* bufs, void* stream) {
+  GruFwdBf16Args p;
+  p.T = ip[0]; p.Emb = ip[1]; p.D = ip[2]; p.H = ip[3];
+  if (p.H % 64) return -1;
+  const uint64_t* b = bufs;
+  p.obs = (const float*)b[0]; p.done = (const uint8_t*)b[1];
+  p.h0 = (const float*)b[2];
+  p.wtin = (const rls_bf16*)b[3]; p.bin = (const float*)b[4];
+  p.wih = (const rls_bf16*)b[5]; p.bih = (const float*)b[6];
+  p.whh = (const rls_bf16*)b[7]; p.bhh = (const float*)b[8];
+  p.xs = (rls_bf16*)b[9]; p.rs = (rls_bf16*)b[10]; p.zs = (rls_bf16*)b[11];
+  p.ns = (rls_bf16*)b[12]; p.ghns = (rls_bf16*)b[13];
+  p.hprevs = (rls_bf16*)b[14]; p.hs = (rls_bf16*)b[15];
+  const int Dp = ((int)p.D + 31) & ~31;
+  size_t lds = sizeof(rls_bf16) * (ROWS * (Dp + 8) + 3 * ROWS * ((int)p.H + 8))
+             + sizeof(float) * 2 * ROWS * (int)p.H;
+  dim3 grid((unsigned)((p.Emb + ROWS - 1) / ROWS));
+  hipLaunchKernelGGL(gru_seq_fwd_bf16_kernel, grid, dim3(256), lds,
+                     (hipStream_t)stream, p);
+  return (int)hipGetLastError();
+}
+
+extern "C" int rlsched_gru_seq_bwd_bf16(const int64_t* ip,
+                                        const uint64_t* bufs, void* stream) {
+  GruBwdBf16Args p;
+  p.T = ip[0]; p.Emb = ip[1]; p.H = ip[2];
+  if (p.H % 64) return -1;
+  const uint64_t* b = bufs;
+  p.dh_ext = (const float*)b[0]; p.done = (const uint8_t*)b[1];
+  p.rs = (const rls_bf16*)b[2]; p.zs = (const rls_bf16*)b[3];
+  p.ns = (const rls_bf16*)b[4]; p.ghns = (const rls_bf16*)b[5];
+  p.hprevs = (const rls_bf16*)b[6]; p.whh_bwd = (const rls_bf16*)b[7];
+  p.dgi_all = (float*)b[8]; p.dgh_all = (float*)b[9];
+  size_t lds = sizeof(rls_bf16) * ROWS * (3 * (int)p.H + 8)
+             + sizeof(float) * 2 * ROWS * ((int)p.H + 4);
+  dim3 grid((unsigned)((p.Emb + ROWS - 1) / ROWS));
+  hipLaunchKernelGGL(gru_seq_bwd_bf16_kernel, grid, dim3(256), lds,
+                     (hipStream_t)stream, p);
+  return (int)hipGetLastError();
+}
+
+extern "C" int rlsched_head_fwd_bf16(const int64_t* ip, const uint64_t* bufs,
+                                     void* stream) {
+  HeadFwdBf16Args p;
+  p.B = ip[0]; p.H = ip[1];
+  if (p.H % 64) return -1;
+  p.h = (const rls_bf16*)bufs[0];
+  p.wth = (const rls_bf16*)bufs[1];
+  p.bh = (const float*)bufs[2];
+  p.head = (float*)bufs[3];
+  size_t lds = sizeof(rls_bf16) * ROWS * ((int)p.H + 8);
+  dim3 grid((unsigned)((p.B + ROWS - 1) / ROWS));
+  hipLaunchKernelGGL(head_fwd_bf16_kernel, grid, dim3(256), lds,
+                     (hipStream_t)stream, p);
+  return (int)hipGetLastError();
+}

@@ -33,3 +33,35 @@ class GruTrainMirror(HipActorGRU):
         self._whT_full[: self.A].copy_(p.pi.weight)
         self._whT_full[self.A].copy_(p.v.weight[0])
         self.whT.copy_(self._pack(self._whT_full))
+
+
+class GruTrainMirrorBf16:
+    """bf16 packed8 mirrors for the mixed-precision GRU BPTT path."""
+
+    def __init__(self, policy):
+        from .hip_actor import HipActorGRUBf16
+        self._fwd = HipActorGRUBf16(policy)
+        H = policy.hidden
+        dev = policy.pi.weight.device
+        bf = torch.bfloat16
+        self.wih_bwd = torch.empty(3 * H // 32, 4, H, 8, device=dev, dtype=bf)
+        self.whh_bwd = torch.empty(3 * H // 32, 4, H, 8, device=dev, dtype=bf)
+        self.whT = torch.empty(32 // 32, 4, H, 8, device=dev, dtype=bf)
+        self._whT_full = torch.zeros(HEADC, H, device=dev)
+        self.policy = policy
+        self.refresh()
+
+    def __getattr__(self, name):
+        if name in ("wtin", "bin", "wih", "bih", "whh", "bhh", "wth", "bh"):
+            return getattr(self._fwd, name)
+        raise AttributeError(name)
+
+    @torch.no_grad()
+    def refresh(self):
+        self._fwd.refresh()
+        p = self.policy
+        self.wih_bwd.copy_(self._fwd._pack8(p.cell.weight_ih.contiguous()))
+        self.whh_bwd.copy_(self._fwd._pack8(p.cell.weight_hh.contiguous()))
+        self._whT_full[: p.num_actions].copy_(p.pi.weight)
+        self._whT_full[p.num_actions].copy_(p.v.weight[0])
+        self.whT.copy_(self._fwd._pack8(self._whT_full))

@@ -64,3 +64,40 @@ def head_fwd(h, wth, bh, head):
     """head[B,32] = h[B,H] @ wth + bh."""
     B, H = h.shape
     _call("rlsched_head_fwd", [B, H], (h, wth, bh, head), _stream(h))
+
+
+def _load_bf16():
+    lib = load()
+    if not hasattr(lib, "_gbf16_ready"):
+        for fn in ("rlsched_gru_seq_fwd_bf16", "rlsched_gru_seq_bwd_bf16",
+                   "rlsched_head_fwd_bf16"):
+            f = getattr(lib, fn)
+            f.restype = ctypes.c_int
+            f.argtypes = [ctypes.c_void_p] * 3
+        lib._gbf16_ready = True
+    return lib
+
+
+def gru_seq_fwd_bf16(obs, done, h0, wtin, bin_, wih, bih, whh, bhh,
+                     xs, rs, zs, ns, ghns, hprevs, hs):
+    _load_bf16()
+    T, Emb, D = obs.shape
+    H = h0.shape[1]
+    _call("rlsched_gru_seq_fwd_bf16", [T, Emb, D, H],
+          (obs, done, h0, wtin, bin_, wih, bih, whh, bhh,
+           xs, rs, zs, ns, ghns, hprevs, hs), _stream(obs))
+
+
+def gru_seq_bwd_bf16(dh_ext, done, rs, zs, ns, ghns, hprevs, whh_bwd,
+                     dgi_all, dgh_all):
+    _load_bf16()
+    T, Emb, H = dh_ext.shape
+    _call("rlsched_gru_seq_bwd_bf16", [T, Emb, H],
+          (dh_ext, done, rs, zs, ns, ghns, hprevs, whh_bwd,
+           dgi_all, dgh_all), _stream(dh_ext))
+
+
+def head_fwd_bf16(h, wth, bh, head):
+    _load_bf16()
+    B, H = h.shape
+    _call("rlsched_head_fwd_bf16", [B, H], (h, wth, bh, head), _stream(head))

@@ -94,3 +94,38 @@ def test_hip_gru_learner_graph_trains():
         assert np.isfinite(v), (k, v)
     for p in pol.parameters():
         assert torch.isfinite(p).all()
+
+
+@pytest.mark.gpu
+def test_gru_bf16_update_close_to_fp32():
+    from helpers import make_buf
+    from rlgpuschedule_amd.agents.ppo import PPOConfig
+    from rlgpuschedule_amd.agents.ppo_gru_hip import HipGRUPPOLearner
+    from rlgpuschedule_amd.models.policy import GRUPolicy
+    torch.manual_seed(6)
+    spec = cluster64()
+    T, E, H = 16, 128, 256
+    pol_a = GRUPolicy(spec.obs_dim, spec.num_actions, H).cuda()
+    pol_b = copy.deepcopy(pol_a)
+    p0 = {n: p.detach().clone() for n, p in pol_a.named_parameters()}
+    cfg = PPOConfig(epochs=1, minibatches=1, norm_reward=False)
+    l32 = HipGRUPPOLearner(pol_a, cfg, use_graph=False, dtype="fp32")
+    l16 = HipGRUPPOLearner(pol_b, cfg, use_graph=False, dtype="bf16")
+    buf = make_buf(T, E, spec, "cuda", seed=13)
+    buf.h0 = torch.randn(E, H, device="cuda") * 0.3
+    l32.update(buf)
+    l16.update(buf)
+    for (n, pa), (_, pb) in zip(pol_a.named_parameters(),
+                                pol_b.named_parameters()):
+        da = (pa - p0[n]).flatten()
+        db = (pb - p0[n]).flatten()
+        cos = torch.nn.functional.cosine_similarity(da, db, dim=0).item()
+        assert cos > 0.98, (n, cos)
+    # graph-mode stability
+    pol_c = GRUPolicy(spec.obs_dim, spec.num_actions, H).cuda()
+    lg = HipGRUPPOLearner(pol_c, PPOConfig(epochs=1, minibatches=2),
+                          use_graph=True, dtype="bf16")
+    for _ in range(3):
+        stats = lg.update(buf)
+    for k, v in stats.items():
+        assert np.isfinite(v), (k, v)
