@@ -104,7 +104,11 @@ def train_impala(cfg, iterations=None, log_fn=print) -> dict:
     policy = make_policy(cfg, device)
     broadcast_params(policy)
     grad_sync = FlatGradAllreduce(policy) if world > 1 else None
-    learner = VtraceLearner(policy, icfg, grad_sync)
+    if device.type == "cuda" and not policy.is_recurrent and cfg.use_hip_env:
+        from .impala_hip import HipVtraceLearner
+        learner = HipVtraceLearner(policy, icfg, grad_sync)
+    else:
+        learner = VtraceLearner(policy, icfg, grad_sync)
     iters = iterations if iterations is not None else cfg.iterations
     stats = {}
 

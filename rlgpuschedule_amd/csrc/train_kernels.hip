@@ -919,3 +919,116 @@ extern "C" int rlsched_bwd_dw_bf16(const int64_t* ip, const uint64_t* bufs,
                      (hipStream_t)stream, p);
   return (int)hipGetLastError();
 }
+
+// ------------------------------------------- IMPALA loss head (fwd+bwd)
+// Per sample: masked softmax over head logits, logp(a), entropy, and the
+// IMPALA objective  L = -pg_adv*logp + 0.5*vf*(v - vs)^2 - ent*H  with
+// dhead written in one pass (pg_adv and vs come from the V-trace scan,
+// which itself consumes logp computed HERE — so the learner calls this
+// kernel twice: pass 1 (logp_only) fills lp_cur for the scan, pass 2
+// computes the loss/grads with the scan's outputs).
+namespace {
+
+struct VtraceHeadArgs {
+  int64_t B, A;
+  int64_t logp_only;
+  float vf_coef, ent_coef, inv_n;
+  const float* head;       // [B,HEADC]
+  const uint8_t* mask;     // [B,A]
+  const int64_t* actions;  // [B]
+  const float* pg_adv;     // [B] (pass 2)
+  const float* vs;         // [B] (pass 2)
+  float* lp_cur;           // [B] out (pass 1) / in unused
+  float* value;            // [B] out (pass 1): V(x) under current params
+  float* dhead;            // [B,HEADC] out (pass 2)
+  float* stats;            // [4]: pi_loss, v_loss, entropy, 0
+};
+
+__global__ __launch_bounds__(256) void vtrace_head_kernel(VtraceHeadArgs a) {
+  const int64_t i0 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int A = (int)a.A;
+  float s_pi = 0.f, s_v = 0.f, s_ent = 0.f;
+  for (int64_t i = i0; i < a.B; i += (int64_t)gridDim.x * blockDim.x) {
+    const float* lg = a.head + i * HEADC;
+    const uint8_t* mk = a.mask + i * A;
+    const int act = (int)a.actions[i];
+    float mx = -3.0e38f;
+    for (int j = 0; j < A; ++j)
+      if (mk[j]) mx = fmaxf(mx, lg[j]);
+    float sum = 0.f;
+    for (int j = 0; j < A; ++j)
+      if (mk[j]) sum += __expf(lg[j] - mx);
+    const float lsum = __logf(sum);
+    const float logp_a = (lg[act] - mx) - lsum;
+    if (a.logp_only) {
+      a.lp_cur[i] = logp_a;
+      a.value[i] = lg[A];
+      continue;
+    }
+    float H = 0.f;
+    for (int j = 0; j < A; ++j)
+      if (mk[j]) {
+        float lpj = (lg[j] - mx) - lsum;
+        H -= __expf(lpj) * lpj;
+      }
+    const float adv = a.pg_adv[i];
+    s_pi += -adv * logp_a;
+    s_ent += H;
+    const float g_logp = -adv * a.inv_n;
+    float* dlg = a.dhead + i * HEADC;
+    for (int j = 0; j < A; ++j) {
+      float g = 0.f;
+      if (mk[j]) {
+        float lpj = (lg[j] - mx) - lsum;
+        float pj = __expf(lpj);
+        g = g_logp * (((j == act) ? 1.f : 0.f) - pj)
+          + a.ent_coef * pj * (lpj + H) * a.inv_n;
+      }
+      dlg[j] = g;
+    }
+    for (int j = A + 1; j < HEADC; ++j) dlg[j] = 0.f;
+    const float verr = lg[A] - a.vs[i];
+    s_v += 0.5f * verr * verr;
+    dlg[A] = a.vf_coef * verr * a.inv_n;
+  }
+  __shared__ float red[3][256 / WAVE];
+  float vals[3] = {s_pi, s_v, s_ent};
+  const int lane = threadIdx.x & (WAVE - 1), w = threadIdx.x >> 6;
+  #pragma unroll
+  for (int k = 0; k < 3; ++k) {
+    float x = vals[k];
+    #pragma unroll
+    for (int off = 32; off > 0; off >>= 1) x += __shfl_xor(x, off, WAVE);
+    if (lane == 0) red[k][w] = x;
+  }
+  __syncthreads();
+  if (threadIdx.x < 3) {
+    float x = 0.f;
+    for (int q = 0; q < 256 / WAVE; ++q) x += red[threadIdx.x][q];
+    atomicAdd(&a.stats[threadIdx.x], x);
+  }
+}
+
+}  // namespace
+
+extern "C" int rlsched_vtrace_head(const int64_t* ip, const double* fp,
+                                   const uint64_t* bufs, void* stream) {
+  VtraceHeadArgs a;
+  a.B = ip[0]; a.A = ip[1]; a.logp_only = ip[2];
+  a.vf_coef = (float)fp[0]; a.ent_coef = (float)fp[1];
+  a.inv_n = (float)fp[2];
+  a.head = (const float*)bufs[0];
+  a.mask = (const uint8_t*)bufs[1];
+  a.actions = (const int64_t*)bufs[2];
+  a.pg_adv = (const float*)bufs[3];
+  a.vs = (const float*)bufs[4];
+  a.lp_cur = (float*)bufs[5];
+  a.value = (float*)bufs[6];
+  a.dhead = (float*)bufs[7];
+  a.stats = (float*)bufs[8];
+  int grid = (int)((a.B + 255) / 256);
+  if (grid > 2048) grid = 2048;
+  hipLaunchKernelGGL(vtrace_head_kernel, dim3(grid), dim3(256), 0,
+                     (hipStream_t)stream, a);
+  return (int)hipGetLastError();
+}

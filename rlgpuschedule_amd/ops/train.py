@@ -242,3 +242,18 @@ def bwd_dw_bf16(X, dY, dW, bc: int = 1024):
     _call("rlsched_bwd_dw_bf16",
           [B, M, N, bc, 1 if X.dtype == torch.bfloat16 else 0],
           _ptrs(X, dY, dW), _stream(dY))
+
+
+def vtrace_head(head, mask, actions, pg_adv, vs, lp_cur, value, dhead,
+                stats, vf_coef, ent_coef, logp_only=False):
+    """IMPALA loss head. Pass 1 (logp_only): fills lp_cur + value for the
+    V-trace scan. Pass 2: loss sums into stats[0..2] + dhead gradient."""
+    lib = load()
+    if not hasattr(lib, "_vth_ready"):
+        lib.rlsched_vtrace_head.restype = ctypes.c_int
+        lib.rlsched_vtrace_head.argtypes = [ctypes.c_void_p] * 4
+        lib._vth_ready = True
+    B, A = mask.shape
+    _call("rlsched_vtrace_head", [B, A, 1 if logp_only else 0],
+          _ptrs(head, mask, actions, pg_adv, vs, lp_cur, value, dhead, stats),
+          _stream(head), fp=[vf_coef, ent_coef, 1.0 / B])
