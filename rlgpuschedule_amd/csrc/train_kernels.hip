@@ -477,19 +477,24 @@ struct GatherArgs {
 };
 
 __global__ __launch_bounds__(256) void gather_kernel(GatherArgs g) {
-  int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  // 32 lanes cooperate per row: the row's float4s load coalesced instead
+  // of one thread walking 21 cachelines serially
+  const int sub = threadIdx.x & 31;
+  int64_t r = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 5;
   if (r >= g.MB) return;
   int64_t s = g.ix[r];
   const int D = (int)g.D, A = (int)g.A;
   const f32x4* src = (const f32x4*)(g.obs + s * D);
   f32x4* dst = (f32x4*)(g.o_obs + r * D);
-  for (int d = 0; d < D / 4; ++d) dst[d] = src[d];
-  for (int a = 0; a < A; ++a) g.o_mask[r * A + a] = g.mask[s * A + a];
-  g.o_acts[r] = g.acts[s];
-  g.o_olp[r] = g.olp[s];
-  g.o_ov[r] = g.ov[s];
-  g.o_adv[r] = g.adv[s];
-  g.o_ret[r] = g.ret[s];
+  for (int d = sub; d < D / 4; d += 32) dst[d] = src[d];
+  for (int a = sub; a < A; a += 32) g.o_mask[r * A + a] = g.mask[s * A + a];
+  if (sub == 0) {
+    g.o_acts[r] = g.acts[s];
+    g.o_olp[r] = g.olp[s];
+    g.o_ov[r] = g.ov[s];
+    g.o_adv[r] = g.adv[s];
+    g.o_ret[r] = g.ret[s];
+  }
 }
 
 }  // namespace
@@ -514,7 +519,7 @@ extern "C" int rlsched_gather_mb(const int64_t* ip, const uint64_t* bufs,
   g.o_ov = (float*)bufs[12];
   g.o_adv = (float*)bufs[13];
   g.o_ret = (float*)bufs[14];
-  dim3 grid((unsigned)((g.MB + 255) / 256));
+  dim3 grid((unsigned)((g.MB * 32 + 255) / 256));
   hipLaunchKernelGGL(gather_kernel, grid, dim3(256), 0, (hipStream_t)stream, g);
   return (int)hipGetLastError();
 }
@@ -755,11 +760,17 @@ __global__ __launch_bounds__(256) void bwd_dx_bf16_kernel(BwdDxBf16Args p) {
   const int lda = K + 8;
   extern __shared__ __attribute__((aligned(16))) char smem[];
   rls_bf16* sdy = (rls_bf16*)smem;    // [ROWS][K+8] plain
-  for (int ix = tid; ix < ROWS * K; ix += 256) {
-    int r = ix / K, k = ix % K;
-    float vv = 0.0f;
-    if (row0 + r < p.B) vv = p.dY[(row0 + r) * K + k];
-    sdy[r * lda + k] = (rls_bf16)vv;
+  // vectorized staging: one float4 dY load + one 8-byte bf16x4 store per
+  // (row, 4-col) slot (K % 4 == 0; contiguous-k LDS layout)
+  typedef rls_bf16 bf16x4 __attribute__((ext_vector_type(4)));
+  for (int ix = tid; ix < ROWS * (K / 4); ix += 256) {
+    int r = ix / (K / 4), k4 = (ix % (K / 4)) * 4;
+    f32x4 vv = {};
+    if (row0 + r < p.B) vv = *(const f32x4*)(p.dY + (row0 + r) * K + k4);
+    bf16x4 bv;
+    #pragma unroll
+    for (int t = 0; t < 4; ++t) bv[t] = (rls_bf16)vv[t];
+    *(bf16x4*)(sdy + r * lda + k4) = bv;
   }
   __syncthreads();
   for (int c0 = w * 64; c0 < N; c0 += 256) {
@@ -815,22 +826,39 @@ __global__ __launch_bounds__(256) void bwd_dw_bf16_kernel(BwdDwBf16Args p) {
   const int wm = w * 16;
   const int ar = lane & 15, kk = lane >> 4, bc = lane & 15;
   for (int kc = 0; kc < (int)p.BC; kc += DW_KC) {
+    typedef rls_bf16 bf16x4 __attribute__((ext_vector_type(4)));
     for (int ix = tid; ix < DW_KC * 16; ix += 256) {
       int k = ix >> 4, j4 = (ix & 15) * 4;
       int64_t b = b0 + kc + k;
-      float xv[4] = {}, yv[4] = {};
+      rls_bf16 xv[4] = {}, yv[4] = {};
       if (b < p.B) {
-        for (int t = 0; t < 4; ++t) {
-          if (m0 + j4 + t < M)
-            xv[t] = p.Xf ? p.Xf[b * M + m0 + j4 + t]
-                         : (float)p.Xb[b * M + m0 + j4 + t];
-          if (n0 + j4 + t < N) yv[t] = p.dY[b * N + n0 + j4 + t];
+        if (m0 + j4 + 3 < M) {
+          if (p.Xf) {
+            f32x4 v = *(const f32x4*)(p.Xf + b * M + m0 + j4);
+            #pragma unroll
+            for (int t = 0; t < 4; ++t) xv[t] = (rls_bf16)v[t];
+          } else {
+            *(bf16x4*)xv = *(const bf16x4*)(p.Xb + b * M + m0 + j4);
+          }
+        } else {
+          for (int t = 0; t < 4; ++t)
+            if (m0 + j4 + t < M)
+              xv[t] = (rls_bf16)(p.Xf ? p.Xf[b * M + m0 + j4 + t]
+                                      : (float)p.Xb[b * M + m0 + j4 + t]);
+        }
+        if (n0 + j4 + 3 < N) {
+          f32x4 v = *(const f32x4*)(p.dY + b * N + n0 + j4);
+          #pragma unroll
+          for (int t = 0; t < 4; ++t) yv[t] = (rls_bf16)v[t];
+        } else {
+          for (int t = 0; t < 4; ++t)
+            if (n0 + j4 + t < N) yv[t] = (rls_bf16)p.dY[b * N + n0 + j4 + t];
         }
       }
       #pragma unroll
       for (int t = 0; t < 4; ++t) {
-        xt[(j4 + t) * DWB_LD + k] = (rls_bf16)xv[t];
-        yt[(j4 + t) * DWB_LD + k] = (rls_bf16)yv[t];
+        xt[(j4 + t) * DWB_LD + k] = xv[t];
+        yt[(j4 + t) * DWB_LD + k] = yv[t];
       }
     }
     __syncthreads();
