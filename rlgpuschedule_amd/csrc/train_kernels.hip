@@ -677,11 +677,24 @@ __global__ __launch_bounds__(256) void fwd_train_bf16_kernel(
   rls_bf16* sobs = (rls_bf16*)smem;
   rls_bf16* sh1 = sobs + ROWS * lda_o;
   rls_bf16* sh2 = sh1 + ROWS * lda_h;
-  for (int ix = tid; ix < ROWS * Dp; ix += 256) {
-    int r = ix / Dp, d = ix % Dp;
-    float vv = 0.0f;
-    if (d < D && row0 + r < p.B) vv = p.obs[(row0 + r) * D + d];
-    sobs[r * lda_o + d] = (rls_bf16)vv;
+  {
+    typedef rls_bf16 bf16x4 __attribute__((ext_vector_type(4)));
+    for (int ix = tid; ix < ROWS * (Dp / 4); ix += 256) {
+      int r = ix / (Dp / 4), d4 = (ix % (Dp / 4)) * 4;
+      bf16x4 bv = {};
+      if (row0 + r < p.B) {
+        if (d4 + 3 < D) {
+          f32x4 v = *(const f32x4*)(p.obs + (row0 + r) * D + d4);
+          #pragma unroll
+          for (int t = 0; t < 4; ++t) bv[t] = (rls_bf16)v[t];
+        } else {
+          for (int t = 0; t < 4; ++t)
+            if (d4 + t < D)
+              bv[t] = (rls_bf16)p.obs[(row0 + r) * D + d4 + t];
+        }
+      }
+      *(bf16x4*)(sobs + r * lda_o + d4) = bv;
+    }
   }
   __syncthreads();
   for (int c0 = w * 64; c0 < H; c0 += 256) {
@@ -807,7 +820,8 @@ struct BwdDwBf16Args {
   float* dW;
 };
 
-#define DWB_LD (DW_KC + 8)   // bf16 elements; /2 dwords % 64 == 4
+#define DWB_KC 256           // bf16 k-chunk (2x f32: LDS is half the bytes)
+#define DWB_LD (DWB_KC + 8)  // bf16 elements; /2 dwords % 64 == 4
 
 __global__ __launch_bounds__(256) void bwd_dw_bf16_kernel(BwdDwBf16Args p) {
   const int tid = threadIdx.x;
@@ -825,9 +839,9 @@ __global__ __launch_bounds__(256) void bwd_dw_bf16_kernel(BwdDwBf16Args p) {
   f32x4 acc[4] = {};
   const int wm = w * 16;
   const int ar = lane & 15, kk = lane >> 4, bc = lane & 15;
-  for (int kc = 0; kc < (int)p.BC; kc += DW_KC) {
+  for (int kc = 0; kc < (int)p.BC; kc += DWB_KC) {
     typedef rls_bf16 bf16x4 __attribute__((ext_vector_type(4)));
-    for (int ix = tid; ix < DW_KC * 16; ix += 256) {
+    for (int ix = tid; ix < DWB_KC * 16; ix += 256) {
       int k = ix >> 4, j4 = (ix & 15) * 4;
       int64_t b = b0 + kc + k;
       rls_bf16 xv[4] = {}, yv[4] = {};
@@ -865,7 +879,7 @@ __global__ __launch_bounds__(256) void bwd_dw_bf16_kernel(BwdDwBf16Args p) {
     const rls_bf16* arow = xt + (wm + ar) * DWB_LD + kk * 8;
     const rls_bf16* brows = yt + bc * DWB_LD + kk * 8;
     #pragma unroll
-    for (int k32 = 0; k32 < DW_KC / 32; ++k32) {
+    for (int k32 = 0; k32 < DWB_KC / 32; ++k32) {
       bf16x8 av = *(const bf16x8*)(arow + 32 * k32);
       #pragma unroll
       for (int n = 0; n < 4; ++n) {
@@ -934,7 +948,7 @@ extern "C" int rlsched_bwd_dw_bf16(const int64_t* ip, const uint64_t* bufs,
   BwdDwBf16Args p;
   p.B = ip[0]; p.M = ip[1]; p.N = ip[2]; p.BC = ip[3];
   const int64_t x_is_bf16 = ip[4];
-  if (p.BC % DW_KC) return -1;
+  if (p.BC % DWB_KC) return -1;
   p.Xf = x_is_bf16 ? nullptr : (const float*)bufs[0];
   p.Xb = x_is_bf16 ? (const rls_bf16*)bufs[0] : nullptr;
   p.dY = (const float*)bufs[1];
