@@ -61,7 +61,8 @@ struct EnvView {
   const int64_t *trace_arr, *trace_work;
   const int32_t *trace_gpus, *trace_ming;
   int64_t job_lo, job_hi, base;
-  int32_t *node_free;
+  int32_t *node_free;        // LDS slice during kernel execution
+  int32_t *node_free_glb;    // global backing (written back at the end)
   int32_t *pend_job;
   int64_t *pend_override;
   int32_t *run_valid, *run_job, *run_gpus;
@@ -397,12 +398,35 @@ __device__ void reset_env(const EnvView& v, Sc& s, int lane) {
   admit(v, s);
 }
 
+// node_free lives in LDS for the kernel's duration: the first-fit scans
+// (placement + K per-slot fits bits every step) are the hottest dependent
+// loads in the transition; LDS turns them from L2 round trips into
+// near-register accesses. Copied in at launch, written back at the end.
+// (Each wave touches only its own slice — no barriers needed.)
+#define RLS_MAX_NODES 128
+
+__device__ __forceinline__ void nf_to_lds(EnvView& v, int32_t* s_nf,
+                                          int lane) {
+  const int N = (int)v.p->N;
+  for (int n = lane; n < N; n += RLS_WAVE) s_nf[n] = v.node_free[n];
+  v.node_free_glb = v.node_free;
+  v.node_free = s_nf;
+}
+
+__device__ __forceinline__ void nf_writeback(const EnvView& v, int lane) {
+  const int N = (int)v.p->N;
+  for (int n = lane; n < N; n += RLS_WAVE) v.node_free_glb[n] = v.node_free[n];
+}
+
 __global__ __launch_bounds__(RLS_BLOCK) void env_reset_kernel(RlsParams p, RlsBufs b) {
+  __shared__ int32_t s_nf[RLS_ENVS_PER_BLOCK][RLS_MAX_NODES];
   int lane = threadIdx.x & (RLS_WAVE - 1);
-  int64_t e = (int64_t)blockIdx.x * RLS_ENVS_PER_BLOCK + (threadIdx.x >> 6);
+  int wid = threadIdx.x >> 6;
+  int64_t e = (int64_t)blockIdx.x * RLS_ENVS_PER_BLOCK + wid;
   if (e >= p.E) return;
   EnvView v = make_view(p, b, e);
   Sc s;
+  nf_to_lds(v, s_nf[wid], lane);
   reset_env(v, s, lane);
   store_sc(b, e, s);
   b.ep_count[e] = 0;
@@ -413,15 +437,19 @@ __global__ __launch_bounds__(RLS_BLOCK) void env_reset_kernel(RlsParams p, RlsBu
   b.rewards[e] = 0.0f;
   b.done[e] = 0;
   write_obs_mask(v, s, lane);
+  nf_writeback(v, lane);
 }
 
 __global__ __launch_bounds__(RLS_BLOCK) void env_step_kernel(RlsParams p, RlsBufs b) {
+  __shared__ int32_t s_nf[RLS_ENVS_PER_BLOCK][RLS_MAX_NODES];
   int lane = threadIdx.x & (RLS_WAVE - 1);
-  int64_t e = (int64_t)blockIdx.x * RLS_ENVS_PER_BLOCK + (threadIdx.x >> 6);
+  int wid = threadIdx.x >> 6;
+  int64_t e = (int64_t)blockIdx.x * RLS_ENVS_PER_BLOCK + wid;
   if (e >= p.E) return;
   EnvView v = make_view(p, b, e);
   Sc s;
   load_sc(b, e, s);
+  nf_to_lds(v, s_nf[wid], lane);
   int64_t a = b.actions[e];
   bool did = false;
   float reward = 0.0f;
@@ -449,6 +477,7 @@ __global__ __launch_bounds__(RLS_BLOCK) void env_step_kernel(RlsParams p, RlsBuf
   }
   store_sc(b, e, s);
   write_obs_mask(v, s, lane);
+  nf_writeback(v, lane);
 }
 
 void unpack(const int64_t* params, const uint64_t* bufs, RlsParams& p, RlsBufs& b) {

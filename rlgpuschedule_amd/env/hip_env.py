@@ -28,6 +28,7 @@ class HipVectorEnv:
             raise RuntimeError("HipVectorEnv needs a GPU (use CpuVectorEnv here)")
         assert spec.queue_visible <= 63, "kernel assumes K <= 63 (one lane/slot)"
         assert spec.max_span <= 8, "RLS_MAX_SPAN in env_common.h is 8"
+        assert spec.num_nodes <= 128, "RLS_MAX_NODES in env_step.hip is 128"
         self.spec = spec
         self.num_envs = E = num_envs
         self.device = torch.device(device)
