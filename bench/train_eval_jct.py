@@ -22,7 +22,7 @@ from rlgpuschedule_amd.trainer import CONFIGS, make_policy  # noqa: E402
 
 def run(iters=300, envs=2048, eval_envs=16, eval_eps=16, jobs=96,
         interarrival=90.0, duration=600.0, seed=7, phase2_iters=0,
-        phase2_ent=0.002, phase2_lr=1e-4, dtype="fp32"):
+        phase2_ent=0.002, phase2_lr=1e-4, dtype="fp32", hidden=256):
     """Programmatic version returning the full comparison dict.
     Optional phase 2: entropy/lr anneal via a fresh learner (Adam state
     resets; the captured graph bakes coefficients, so annealing means a
@@ -37,7 +37,9 @@ def run(iters=300, envs=2048, eval_envs=16, eval_eps=16, jobs=96,
     torch.manual_seed(seed)
     env = HipVectorEnv(spec, envs, trace_kind="philly", jobs_per_env=jobs,
                        seed=seed, **tkw)
-    policy = make_policy(CONFIGS["c2_ppo_mlp"], torch.device("cuda"))
+    import dataclasses
+    cfg0 = dataclasses.replace(CONFIGS["c2_ppo_mlp"], hidden=hidden)
+    policy = make_policy(cfg0, torch.device("cuda"))
     learner = HipPPOLearner(policy, PPOConfig(epochs=1, minibatches=4), dtype=dtype)
     engine = GpuRolloutEngine(env, policy, 64, seed=seed)
     t0 = time.perf_counter()
