@@ -22,7 +22,8 @@ from rlgpuschedule_amd.trainer import CONFIGS, make_policy  # noqa: E402
 
 def run(iters=300, envs=2048, eval_envs=16, eval_eps=16, jobs=96,
         interarrival=90.0, duration=600.0, seed=7, phase2_iters=0,
-        phase2_ent=0.002, phase2_lr=1e-4, dtype="fp32", hidden=256):
+        phase2_ent=0.002, phase2_lr=1e-4, dtype="fp32", hidden=256,
+        save_ckpt=None):
     """Programmatic version returning the full comparison dict.
     Optional phase 2: entropy/lr anneal via a fresh learner (Adam state
     resets; the captured graph bakes coefficients, so annealing means a
@@ -57,6 +58,10 @@ def run(iters=300, envs=2048, eval_envs=16, eval_eps=16, jobs=96,
     torch.cuda.synchronize()
     res = {"train_s": round(time.perf_counter() - t0, 1),
            "final": {k: round(v, 4) for k, v in stats.items()}}
+    if save_ckpt:
+        from rlgpuschedule_amd.ckpt.io import save_policy
+        save_policy(save_ckpt, policy, spec=spec,
+                    meta={"iters": iters, "hidden": hidden, "dtype": dtype})
     ekw = dict(num_envs=eval_envs, min_episodes=eval_eps,
                jobs_per_env=jobs, seed=991, **tkw)
     res["ppo_greedy"] = evaluate_policy(spec, policy, greedy=True,
