@@ -23,7 +23,7 @@ from rlgpuschedule_amd.trainer import CONFIGS, make_policy  # noqa: E402
 def run(iters=300, envs=2048, eval_envs=16, eval_eps=16, jobs=96,
         interarrival=90.0, duration=600.0, seed=7, phase2_iters=0,
         phase2_ent=0.002, phase2_lr=1e-4, dtype="fp32", hidden=256,
-        save_ckpt=None):
+        save_ckpt=None, ent_coef=0.01, lr=3e-4):
     """Programmatic version returning the full comparison dict.
     Optional phase 2: entropy/lr anneal via a fresh learner (Adam state
     resets; the captured graph bakes coefficients, so annealing means a
@@ -41,7 +41,9 @@ def run(iters=300, envs=2048, eval_envs=16, eval_eps=16, jobs=96,
     import dataclasses
     cfg0 = dataclasses.replace(CONFIGS["c2_ppo_mlp"], hidden=hidden)
     policy = make_policy(cfg0, torch.device("cuda"))
-    learner = HipPPOLearner(policy, PPOConfig(epochs=1, minibatches=4), dtype=dtype)
+    learner = HipPPOLearner(policy, PPOConfig(epochs=1, minibatches=4,
+                                              ent_coef=ent_coef, lr=lr),
+                            dtype=dtype)
     engine = GpuRolloutEngine(env, policy, 64, seed=seed)
     t0 = time.perf_counter()
     stats = {}
