@@ -25,6 +25,8 @@ def main(argv=None):
     tr.add_argument("--out", default=None, help="metrics/ckpt directory")
     tr.add_argument("--save-every", type=int, default=0)
     tr.add_argument("--resume", default=None, help="checkpoint path")
+    tr.add_argument("--actor-dtype", choices=["fp32", "bf16"], default=None)
+    tr.add_argument("--update-dtype", choices=["fp32", "bf16"], default=None)
     ev = sub.add_parser("eval", help="evaluate a checkpoint or baseline")
     ev.add_argument("--config", choices=sorted(CONFIGS), default="c2_ppo_mlp")
     ev.add_argument("--ckpt", default=None, help="policy checkpoint (.pt/.safetensors)")
@@ -82,6 +84,10 @@ def main(argv=None):
         over["device"] = args.device
     if args.jobs_per_env is not None:
         over["jobs_per_env"] = args.jobs_per_env
+    if args.actor_dtype is not None:
+        over["actor_dtype"] = args.actor_dtype
+    if args.update_dtype is not None:
+        over["update_dtype"] = args.update_dtype
     if over:
         cfg = dataclasses.replace(cfg, **over)
     out = train(cfg, iterations=args.iterations, out_dir=args.out,

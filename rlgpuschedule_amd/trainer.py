@@ -34,6 +34,8 @@ class TrainConfig:
     a2c: A2CConfig = field(default_factory=A2CConfig)
     use_hip_env: bool = True      # False -> CPU reference env
     trace_kw: dict = field(default_factory=dict)
+    actor_dtype: str = "bf16"     # rollout policy compute (fp32 | bf16)
+    update_dtype: str = "fp32"    # learner GEMMs (fp32 exact | bf16 mixed)
 
 
 CONFIGS = {
@@ -114,7 +116,8 @@ def train(cfg: TrainConfig, iterations: int | None = None,
     if on_gpu_fast:
         from .agents.rollout_gpu import GpuRolloutEngine
         collector = GpuRolloutEngine(env, policy, cfg.rollout_T,
-                                     seed=cfg.seed * 101 + rank)
+                                     seed=cfg.seed * 101 + rank,
+                                     actor_dtype=cfg.actor_dtype)
     else:
         collector = Collector(env, policy, cfg.rollout_T, device,
                               seed=cfg.seed * 101 + rank)
@@ -122,10 +125,12 @@ def train(cfg: TrainConfig, iterations: int | None = None,
         learner = None
     elif (on_gpu_fast and cfg.algo == "ppo" and not policy.is_recurrent):
         from .agents.ppo_hip import HipPPOLearner
-        learner = HipPPOLearner(policy, cfg.ppo, grad_sync)
+        learner = HipPPOLearner(policy, cfg.ppo, grad_sync,
+                                dtype=cfg.update_dtype)
     elif on_gpu_fast and cfg.algo == "ppo" and policy.is_recurrent:
         from .agents.ppo_gru_hip import HipGRUPPOLearner
-        learner = HipGRUPPOLearner(policy, cfg.ppo, grad_sync)
+        learner = HipGRUPPOLearner(policy, cfg.ppo, grad_sync,
+                                   dtype=cfg.update_dtype)
     else:
         learner = make_learner(cfg, policy, grad_sync)
     start_it = 0
