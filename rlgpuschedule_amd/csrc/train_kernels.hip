@@ -132,7 +132,13 @@ __global__ __launch_bounds__(256) void ppo_head_kernel(PpoHeadArgs a) {
   const int A = (int)a.A;
   float s_pi = 0.f, s_v = 0.f, s_ent = 0.f, s_cf = 0.f;
   for (int64_t i = i0; i < a.B; i += (int64_t)gridDim.x * blockDim.x) {
-    const float* lg = a.head + i * HEADC;
+    // pull the whole head row (128 B) into registers with 8 f32x4 loads
+    // and the mask with 4 dword loads — one pass over memory per sample
+    float lgr[HEADC];
+    #pragma unroll
+    for (int q = 0; q < HEADC / 4; ++q)
+      *(f32x4*)(lgr + 4 * q) = *(const f32x4*)(a.head + i * HEADC + 4 * q);
+    const float* lg = lgr;
     const uint8_t* mk = a.mask + i * A;
     float* dlg = a.dhead + i * HEADC;
     const int act = (int)a.actions[i];
@@ -342,20 +348,23 @@ struct ColSumArgs {
 };
 
 __global__ __launch_bounds__(256) void colsum_kernel(ColSumArgs p) {
-  int64_t n = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  if (n >= p.N) return;
+  // thread owns FOUR consecutive columns and reads f32x4 per row:
+  // consecutive threads read consecutive 16B (fully coalesced); N % 4
+  // required (all grad widths are multiples of 4)
+  int64_t n4 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 4;
+  if (n4 >= p.N) return;
   int64_t b0 = (int64_t)blockIdx.y * p.BCS;
   int64_t b1 = b0 + p.BCS < p.B ? b0 + p.BCS : p.B;
-  float s0 = 0.f, s1 = 0.f, s2 = 0.f, s3 = 0.f;
+  f32x4 s0 = {}, s1 = {};
   int64_t b = b0;
-  for (; b + 3 < b1; b += 4) {   // 4 independent loads in flight
-    s0 += p.dY[b * p.N + n];
-    s1 += p.dY[(b + 1) * p.N + n];
-    s2 += p.dY[(b + 2) * p.N + n];
-    s3 += p.dY[(b + 3) * p.N + n];
+  for (; b + 1 < b1; b += 2) {   // 2 independent f32x4 loads in flight
+    s0 += *(const f32x4*)(p.dY + b * p.N + n4);
+    s1 += *(const f32x4*)(p.dY + (b + 1) * p.N + n4);
   }
-  for (; b < b1; ++b) s0 += p.dY[b * p.N + n];
-  atomicAdd(&p.db[n], (s0 + s1) + (s2 + s3));
+  for (; b < b1; ++b) s0 += *(const f32x4*)(p.dY + b * p.N + n4);
+  s0 += s1;
+  #pragma unroll
+  for (int t = 0; t < 4; ++t) atomicAdd(&p.db[n4 + t], s0[t]);
 }
 
 }  // namespace
