@@ -132,13 +132,10 @@ __global__ __launch_bounds__(256) void ppo_head_kernel(PpoHeadArgs a) {
   const int A = (int)a.A;
   float s_pi = 0.f, s_v = 0.f, s_ent = 0.f, s_cf = 0.f;
   for (int64_t i = i0; i < a.B; i += (int64_t)gridDim.x * blockDim.x) {
-    // pull the whole head row (128 B) into registers with 8 f32x4 loads
-    // and the mask with 4 dword loads — one pass over memory per sample
-    float lgr[HEADC];
-    #pragma unroll
-    for (int q = 0; q < HEADC / 4; ++q)
-      *(f32x4*)(lgr + 4 * q) = *(const f32x4*)(a.head + i * HEADC + 4 * q);
-    const float* lg = lgr;
+    // NOTE: do NOT stage the row into a local array — runtime-indexed
+    // per-thread arrays go to scratch (5x); the row is L2/L1-resident
+    // across the four passes anyway
+    const float* lg = a.head + i * HEADC;
     const uint8_t* mk = a.mask + i * A;
     float* dlg = a.dhead + i * HEADC;
     const int act = (int)a.actions[i];
