@@ -344,7 +344,7 @@ struct ColSumArgs {
   float* db;
 };
 
-__global__ __launch_bounds__(256) void colsum_kernel(ColSumArgs p) {
+__global__ __launch_bounds__(64) void colsum_kernel(ColSumArgs p) {
   // thread owns FOUR consecutive columns and reads f32x4 per row:
   // consecutive threads read consecutive 16B (fully coalesced); N % 4
   // required (all grad widths are multiples of 4)
