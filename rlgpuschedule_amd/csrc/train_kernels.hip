@@ -448,12 +448,15 @@ extern "C" int rlsched_colsum(const int64_t* ip, const uint64_t* bufs,
                               void* stream) {
   ColSumArgs p;
   p.B = ip[0]; p.N = ip[1];
+  if (p.N % 4) return -1;
   p.BCS = 128;
   p.dY = (const float*)bufs[0];
   p.db = (float*)bufs[1];
   unsigned chunks = (unsigned)((p.B + p.BCS - 1) / p.BCS);
-  dim3 grid((unsigned)((p.N + 255) / 256), chunks);
-  hipLaunchKernelGGL(colsum_kernel, grid, dim3(256), 0, (hipStream_t)stream, p);
+  // one WAVE per (64-column tile, chunk): every lane active (a 256-thread
+  // block would idle 75% at N=256)
+  dim3 grid((unsigned)((p.N / 4 + 63) / 64), chunks);
+  hipLaunchKernelGGL(colsum_kernel, grid, dim3(64), 0, (hipStream_t)stream, p);
   return (int)hipGetLastError();
 }
 
