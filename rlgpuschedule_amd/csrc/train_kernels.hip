@@ -344,24 +344,21 @@ struct ColSumArgs {
   float* db;
 };
 
-__global__ __launch_bounds__(64) void colsum_kernel(ColSumArgs p) {
-  // thread owns FOUR consecutive columns and reads f32x4 per row:
-  // consecutive threads read consecutive 16B (fully coalesced); N % 4
-  // required (all grad widths are multiples of 4)
-  int64_t n4 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 4;
-  if (n4 >= p.N) return;
+__global__ __launch_bounds__(256) void colsum_kernel(ColSumArgs p) {
+  int64_t n = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (n >= p.N) return;
   int64_t b0 = (int64_t)blockIdx.y * p.BCS;
   int64_t b1 = b0 + p.BCS < p.B ? b0 + p.BCS : p.B;
-  f32x4 s0 = {}, s1 = {};
+  float s0 = 0.f, s1 = 0.f, s2 = 0.f, s3 = 0.f;
   int64_t b = b0;
-  for (; b + 1 < b1; b += 2) {   // 2 independent f32x4 loads in flight
-    s0 += *(const f32x4*)(p.dY + b * p.N + n4);
-    s1 += *(const f32x4*)(p.dY + (b + 1) * p.N + n4);
+  for (; b + 3 < b1; b += 4) {   // 4 independent loads in flight
+    s0 += p.dY[b * p.N + n];
+    s1 += p.dY[(b + 1) * p.N + n];
+    s2 += p.dY[(b + 2) * p.N + n];
+    s3 += p.dY[(b + 3) * p.N + n];
   }
-  for (; b < b1; ++b) s0 += *(const f32x4*)(p.dY + b * p.N + n4);
-  s0 += s1;
-  #pragma unroll
-  for (int t = 0; t < 4; ++t) atomicAdd(&p.db[n4 + t], s0[t]);
+  for (; b < b1; ++b) s0 += p.dY[b * p.N + n];
+  atomicAdd(&p.db[n], (s0 + s1) + (s2 + s3));
 }
 
 }  // namespace
@@ -448,15 +445,12 @@ extern "C" int rlsched_colsum(const int64_t* ip, const uint64_t* bufs,
                               void* stream) {
   ColSumArgs p;
   p.B = ip[0]; p.N = ip[1];
-  if (p.N % 4) return -1;
   p.BCS = 128;
   p.dY = (const float*)bufs[0];
   p.db = (float*)bufs[1];
   unsigned chunks = (unsigned)((p.B + p.BCS - 1) / p.BCS);
-  // one WAVE per (64-column tile, chunk): every lane active (a 256-thread
-  // block would idle 75% at N=256)
-  dim3 grid((unsigned)((p.N / 4 + 63) / 64), chunks);
-  hipLaunchKernelGGL(colsum_kernel, grid, dim3(64), 0, (hipStream_t)stream, p);
+  dim3 grid((unsigned)((p.N + 255) / 256), chunks);
+  hipLaunchKernelGGL(colsum_kernel, grid, dim3(256), 0, (hipStream_t)stream, p);
   return (int)hipGetLastError();
 }
 
