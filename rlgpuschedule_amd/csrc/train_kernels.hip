@@ -445,7 +445,7 @@ extern "C" int rlsched_colsum(const int64_t* ip, const uint64_t* bufs,
                               void* stream) {
   ColSumArgs p;
   p.B = ip[0]; p.N = ip[1];
-  p.BCS = 128;
+  p.BCS = 512;
   p.dY = (const float*)bufs[0];
   p.db = (float*)bufs[1];
   unsigned chunks = (unsigned)((p.B + p.BCS - 1) / p.BCS);
