@@ -23,11 +23,19 @@ from rlgpuschedule_amd.trainer import CONFIGS, make_policy  # noqa: E402
 def run(iters=300, envs=2048, eval_envs=16, eval_eps=16, jobs=96,
         interarrival=90.0, duration=600.0, seed=7, phase2_iters=0,
         phase2_ent=0.002, phase2_lr=1e-4, dtype="fp32", hidden=256,
-        save_ckpt=None, ent_coef=0.01, lr=3e-4):
+        save_ckpt=None, ent_coef=0.01, lr=3e-4, select_every=0,
+        select_envs=8, select_eps=8):
     """Programmatic version returning the full comparison dict.
     Optional phase 2: entropy/lr anneal via a fresh learner (Adam state
     resets; the captured graph bakes coefficients, so annealing means a
-    new capture)."""
+    new capture).
+    select_every>0 enables eval-based checkpoint selection: every
+    select_every iters run a quick greedy eval on a SELECTION trace pool
+    (seed 555, disjoint from the final-eval seed 991 so the reported
+    numbers carry no selection bias) and keep the best-JCT snapshot;
+    the final evaluation then uses that snapshot. This is the standard
+    answer to high-variance RL training (atomic-accumulation
+    nondeterminism makes late training chaotic; see docs/PERF.md)."""
     from rlgpuschedule_amd.agents.ppo import PPOConfig
     from rlgpuschedule_amd.agents.ppo_hip import HipPPOLearner
     from rlgpuschedule_amd.agents.rollout_gpu import GpuRolloutEngine
@@ -47,9 +55,23 @@ def run(iters=300, envs=2048, eval_envs=16, eval_eps=16, jobs=96,
     engine = GpuRolloutEngine(env, policy, 64, seed=seed)
     t0 = time.perf_counter()
     stats = {}
+    best = {"jct": float("inf"), "state": None, "at": -1}
+    skw = dict(num_envs=select_envs, min_episodes=select_eps,
+               jobs_per_env=jobs, seed=555, **tkw)
+
+    def _select(it):
+        st = evaluate_policy(spec, policy, greedy=True, device="cuda", **skw)
+        if st["mean_jct_us"] < best["jct"]:
+            best["jct"] = st["mean_jct_us"]
+            best["state"] = {k: v.detach().clone()
+                             for k, v in policy.state_dict().items()}
+            best["at"] = it
+
     for it in range(iters):
         buf = engine.collect()
         stats = learner.update(buf)
+        if select_every and (it + 1) % select_every == 0:
+            _select(it + 1)
     if phase2_iters:
         learner = HipPPOLearner(
             policy, PPOConfig(epochs=1, minibatches=4, lr=phase2_lr,
@@ -60,6 +82,10 @@ def run(iters=300, envs=2048, eval_envs=16, eval_eps=16, jobs=96,
     torch.cuda.synchronize()
     res = {"train_s": round(time.perf_counter() - t0, 1),
            "final": {k: round(v, 4) for k, v in stats.items()}}
+    if select_every and best["state"] is not None:
+        policy.load_state_dict(best["state"])
+        res["selected_at"] = best["at"]
+        res["selection_jct_us"] = round(best["jct"], 1)
     if save_ckpt:
         from rlgpuschedule_amd.ckpt.io import save_policy
         save_policy(save_ckpt, policy, spec=spec,
