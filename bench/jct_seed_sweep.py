@@ -10,8 +10,9 @@ import sys
 from pathlib import Path
 
 sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
+sys.path.insert(0, str(Path(__file__).resolve().parent))
 
-from bench.train_eval_jct import run  # noqa: E402
+from train_eval_jct import run  # noqa: E402
 
 if __name__ == "__main__":
     seeds = [int(s) for s in (sys.argv[1:] or ["7", "21", "42", "1234"])]
