@@ -61,7 +61,9 @@ def run(iters=300, envs=2048, eval_envs=16, eval_eps=16, jobs=96,
 
     def _select(it):
         st = evaluate_policy(spec, policy, greedy=True, device="cuda", **skw)
-        if st["mean_jct_us"] < best["jct"]:
+        # guard: a degenerate never-schedule policy completes 0 jobs and
+        # reports mean_jct_us == 0 — that must never win selection
+        if st["completed"] > 0 and st["mean_jct_us"] < best["jct"]:
             best["jct"] = st["mean_jct_us"]
             best["state"] = {k: v.detach().clone()
                              for k, v in policy.state_dict().items()}
