@@ -39,6 +39,11 @@ class PPOConfig:
     max_grad_norm: float = 0.5
     norm_adv: bool = True
     norm_reward: bool = True
+    # KL-to-behavior penalty (k3 estimator: ratio-1-log ratio, always >=0).
+    # 0.0 = vanilla PPO. Non-zero counteracts the late-training entropy
+    # collapse documented in docs/PERF.md (policy drifts far off-behavior
+    # between graph-captured updates).
+    kl_coef: float = 0.0
 
 
 class PPOLearner:
@@ -73,6 +78,9 @@ class PPOLearner:
         s1 = ratio * adv
         s2 = ratio.clamp(1.0 - cfg.clip, 1.0 + cfg.clip) * adv
         pi_loss = -torch.min(s1, s2).mean()
+        if cfg.kl_coef:
+            pi_loss = pi_loss + cfg.kl_coef * (ratio - 1.0
+                                               - (logp - old_logp)).mean()
         v_clip = old_value + (value - old_value).clamp(-cfg.clip, cfg.clip)
         v_loss = 0.5 * torch.max((value - ret) ** 2, (v_clip - ret) ** 2).mean()
         ent = entropy.mean()

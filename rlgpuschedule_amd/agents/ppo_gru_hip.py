@@ -189,7 +189,8 @@ class HipGRUPPOLearner:
                          s["acts"].view(B), s["olp"].view(B),
                          s["ov"].view(B), s["advm"].reshape(B),
                          s["retm"].view(B), cfg.clip, cfg.vf_coef,
-                         cfg.ent_coef, s["dhead"], s["stats"])
+                         cfg.ent_coef, s["dhead"], s["stats"],
+                         kl_coef=getattr(cfg, "kl_coef", 0.0))
                 if self.dtype == "bf16":
                     bwd_dx_bf16(s["dhead"], mi.whT, None,
                                 s["dh_ext"].view(B, H))

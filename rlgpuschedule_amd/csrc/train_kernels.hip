@@ -115,7 +115,7 @@ __global__ __launch_bounds__(256) void fwd_train_kernel(FwdTrainArgs p) {
 // gradient dhead [B,HEADC] (zeros for masked/pad columns).
 struct PpoHeadArgs {
   int64_t B, A;
-  float clip, vf_coef, ent_coef, inv_n;
+  float clip, vf_coef, ent_coef, inv_n, kl_coef;
   const float* head;       // [B,HEADC]
   const uint8_t* mask;     // [B,A]
   const int64_t* actions;  // [B]
@@ -158,10 +158,14 @@ __global__ __launch_bounds__(256) void ppo_head_kernel(PpoHeadArgs a) {
     const float rcl = fminf(fmaxf(ratio, 1.f - a.clip), 1.f + a.clip);
     const float s1 = ratio * advi, s2 = rcl * advi;
     const float active = (s1 <= s2) ? 1.f : 0.f;
-    s_pi += -fminf(s1, s2);
+    // k3 KL-to-behavior penalty (ratio-1-log ratio >= 0); kl_coef=0 is
+    // vanilla PPO. Folded into the pi stat; grad wrt logp is ratio-1.
+    const float lr3 = logp_a - a.old_logp[i];
+    s_pi += -fminf(s1, s2) + a.kl_coef * (ratio - 1.f - lr3);
     s_ent += H;
     s_cf += (fabsf(ratio - 1.f) > a.clip) ? 1.f : 0.f;
-    const float g_logp = -advi * ratio * active * a.inv_n;
+    const float g_logp = (-advi * ratio * active
+                          + a.kl_coef * (ratio - 1.f)) * a.inv_n;
     for (int j = 0; j < A; ++j) {
       float g = 0.f;
       if (mk[j]) {
@@ -392,6 +396,7 @@ extern "C" int rlsched_ppo_head(const int64_t* ip, const double* fp,
   a.B = ip[0]; a.A = ip[1];
   a.clip = (float)fp[0]; a.vf_coef = (float)fp[1];
   a.ent_coef = (float)fp[2]; a.inv_n = (float)fp[3];
+  a.kl_coef = (float)fp[4];
   a.head = (const float*)bufs[0];
   a.mask = (const uint8_t*)bufs[1];
   a.actions = (const int64_t*)bufs[2];
