@@ -18,8 +18,10 @@ if __name__ == "__main__":
     seeds = [int(s) for s in (sys.argv[1:] or ["7", "21", "42", "1234"])]
     out = {"runs": []}
     for i, sd in enumerate(seeds):
+        import os
+        kl = float(os.environ.get("SWEEP_KL", "0.0"))
         r = run(iters=3000, envs=2048, hidden=512, seed=sd,
-                select_every=200, dtype="bf16")
+                select_every=200, dtype="bf16", kl_coef=kl)
         rec = {"seed": sd, "train_s": r["train_s"],
                "selected_at": r.get("selected_at"),
                "selection_jct_us": r.get("selection_jct_us"),

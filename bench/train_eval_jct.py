@@ -24,7 +24,7 @@ def run(iters=300, envs=2048, eval_envs=16, eval_eps=16, jobs=96,
         interarrival=90.0, duration=600.0, seed=7, phase2_iters=0,
         phase2_ent=0.002, phase2_lr=1e-4, dtype="fp32", hidden=256,
         save_ckpt=None, ent_coef=0.01, lr=3e-4, select_every=0,
-        select_envs=8, select_eps=8):
+        select_envs=8, select_eps=8, kl_coef=0.0):
     """Programmatic version returning the full comparison dict.
     Optional phase 2: entropy/lr anneal via a fresh learner (Adam state
     resets; the captured graph bakes coefficients, so annealing means a
@@ -50,14 +50,17 @@ def run(iters=300, envs=2048, eval_envs=16, eval_eps=16, jobs=96,
     cfg0 = dataclasses.replace(CONFIGS["c2_ppo_mlp"], hidden=hidden)
     policy = make_policy(cfg0, torch.device("cuda"))
     learner = HipPPOLearner(policy, PPOConfig(epochs=1, minibatches=4,
-                                              ent_coef=ent_coef, lr=lr),
+                                              ent_coef=ent_coef, lr=lr,
+                                              kl_coef=kl_coef),
                             dtype=dtype)
     engine = GpuRolloutEngine(env, policy, 64, seed=seed)
     t0 = time.perf_counter()
     stats = {}
     best = {"jct": float("inf"), "state": None, "at": -1}
+    # max_steps bounds the cost of evaluating a degenerate (never-schedule)
+    # policy: episodes then never finish and the eval would run 200k steps
     skw = dict(num_envs=select_envs, min_episodes=select_eps,
-               jobs_per_env=jobs, seed=555, **tkw)
+               jobs_per_env=jobs, seed=555, max_steps=30_000, **tkw)
 
     def _select(it):
         st = evaluate_policy(spec, policy, greedy=True, device="cuda", **skw)
