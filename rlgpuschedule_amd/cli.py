@@ -38,6 +38,18 @@ def main(argv=None):
     ev.add_argument("--seed", type=int, default=1234)
     ev.add_argument("--sample", action="store_true",
                     help="sample instead of greedy argmax")
+    sw = sub.add_parser("train-sweep", help="multi-seed training with "
+                        "eval-based checkpoint selection (agents/selection)")
+    sw.add_argument("--config", choices=sorted(CONFIGS), default="c2_ppo_mlp")
+    sw.add_argument("--seeds", type=int, nargs="+", default=[7, 21, 42])
+    sw.add_argument("--iterations", type=int, default=None)
+    sw.add_argument("--envs", type=int, default=None)
+    sw.add_argument("--jobs-per-env", type=int, default=None)
+    sw.add_argument("--select-every", type=int, default=200)
+    sw.add_argument("--kl-coef", type=float, default=0.2,
+                    help="KL-to-behavior penalty (prevents collapse)")
+    sw.add_argument("--save-best", default=None, help="checkpoint path for "
+                    "the winning seed's selected policy")
     ls = sub.add_parser("configs", help="list configs")
     args = ap.parse_args(argv)
 
@@ -70,6 +82,26 @@ def main(argv=None):
                                  seed=args.seed, greedy=not args.sample,
                                  **cfg.trace_kw)
         print(json.dumps(st))
+        return 0
+
+    if args.cmd == "train-sweep":
+        import json
+        from .agents.selection import train_multi_seed
+        cfg = CONFIGS[args.config]
+        over = {}
+        if args.envs is not None:
+            over["num_envs"] = args.envs
+        if args.jobs_per_env is not None:
+            over["jobs_per_env"] = args.jobs_per_env
+        if over:
+            cfg = dataclasses.replace(cfg, **over)
+        cfg = dataclasses.replace(
+            cfg, ppo=dataclasses.replace(cfg.ppo, kl_coef=args.kl_coef))
+        res = train_multi_seed(cfg, args.seeds, iterations=args.iterations,
+                               select_every=args.select_every,
+                               log_fn=print, save_best=args.save_best)
+        res.pop("best_state", None)
+        print(json.dumps(res))
         return 0
 
     cfg = CONFIGS[args.config]

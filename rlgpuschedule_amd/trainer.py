@@ -98,8 +98,11 @@ def make_learner(cfg: TrainConfig, policy, grad_sync=None):
 
 def train(cfg: TrainConfig, iterations: int | None = None,
           log_fn=print, out_dir: str | None = None, save_every: int = 0,
-          resume: str | None = None) -> dict:
-    """Synchronous trainer (ppo/a2c). IMPALA lives in agents/impala.py."""
+          resume: str | None = None, callback=None) -> dict:
+    """Synchronous trainer (ppo/a2c). IMPALA lives in agents/impala.py.
+    callback(it, policy, stats), if given, runs on rank 0 after every
+    update (it must not mutate the policy) — used by agents/selection.py
+    for eval-based checkpoint selection."""
     rank, world, local = init_from_env()
     device = resolve_device(cfg)
     if device.type == "cuda":
@@ -155,6 +158,8 @@ def train(cfg: TrainConfig, iterations: int | None = None,
         if learner is not None:
             stats = learner.update(buf)
             check_finite(stats, where=f"iter {it}")
+        if callback is not None and rank == 0:
+            callback(it, policy, stats)
         if logger is not None:
             logger.log(stats, iteration=it)
         if rank == 0 and log_fn and (it % 10 == 0 or it == start_it + iters - 1):
@@ -168,6 +173,7 @@ def train(cfg: TrainConfig, iterations: int | None = None,
                         spec=cfg.spec, optimizer=getattr(learner, "opt", None),
                         meta={"iteration": it + 1, "config": cfg.name})
     out = dict(stats)
+    out["_policy"] = policy   # for selection/eval pipelines; not serialized
     if hasattr(env, "episode_stats"):
         out["episode_stats"] = env.episode_stats()
     elif hasattr(env, "venv"):

@@ -139,3 +139,46 @@ def test_trainer_config2_smoke_cpu():
         use_hip_env=False, device="cpu", hidden=32)
     out = train(cfg, iterations=2, log_fn=None)
     assert np.isfinite(out["pi_loss"])
+
+
+def test_multi_seed_selection_cpu():
+    """train_multi_seed: selection snapshots, disjoint held-out eval, and
+    winner chosen by selection-pool JCT (never the held-out number)."""
+    import dataclasses
+
+    from rlgpuschedule_amd.agents.selection import train_multi_seed
+    from rlgpuschedule_amd.trainer import CONFIGS
+    cfg = dataclasses.replace(
+        CONFIGS["c2_ppo_mlp"], num_envs=4, jobs_per_env=8, rollout_T=8,
+        use_hip_env=False, device="cpu", hidden=16)
+    cfg = dataclasses.replace(
+        cfg, ppo=dataclasses.replace(cfg.ppo, epochs=1, minibatches=1,
+                                     kl_coef=0.2))
+    res = train_multi_seed(cfg, [1, 2], iterations=4, select_every=2,
+                           select_kw=dict(num_envs=2, min_episodes=2,
+                                          max_steps=3000),
+                           eval_kw=dict(num_envs=2, min_episodes=2,
+                                        max_steps=3000))
+    assert len(res["runs"]) == 2
+    for r in res["runs"]:
+        assert r["selected_at"] > 0
+        assert r["heldout_jct_us"] > 0
+    assert res["best_seed"] in (1, 2)
+    win = min(res["runs"], key=lambda r: r["selection_jct_us"])
+    assert res["best_seed"] == win["seed"]
+    assert res["best_selection_jct_us"] == win["selection_jct_us"]
+    # best_state loads into a fresh policy
+    import torch
+
+    from rlgpuschedule_amd.trainer import make_policy
+    pol = make_policy(cfg, torch.device("cpu"))
+    pol.load_state_dict(res["best_state"])
+
+
+def test_cli_train_sweep_cpu(tmp_path):
+    from rlgpuschedule_amd.cli import main
+    rc = main(["train-sweep", "--config", "c2_ppo_mlp", "--seeds", "1",
+               "--iterations", "2", "--envs", "2", "--jobs-per-env", "8",
+               "--select-every", "1",
+               "--save-best", str(tmp_path / "best.pt")])
+    assert rc == 0
