@@ -167,6 +167,9 @@ class HipVtraceLearner:
                 warnings.warn(f"hipGraph capture of the IMPALA update failed "
                               f"({e!r}); falling back to eager execution")
                 self.use_graph = False
+            from ..parallel.dist import agree_all
+            if not agree_all(self.use_graph):   # rank-consistent fallback
+                self.use_graph = False
         if self.use_graph:
             self._graphs[key].replay()
         else:

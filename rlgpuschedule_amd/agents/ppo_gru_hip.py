@@ -269,6 +269,10 @@ class HipGRUPPOLearner:
                               f"({e!r}); falling back to eager execution")
                 self.use_graph = False
                 self._graph = None
+            from ..parallel.dist import agree_all
+            if not agree_all(self.use_graph):   # rank-consistent fallback
+                self.use_graph = False
+                self._graph = None
         if self.use_graph:
             self._graph.replay()
         else:

@@ -241,6 +241,12 @@ class HipPPOLearner:
                               f"({e!r}); falling back to eager execution")
                 self.use_graph = False
                 self._graph = None
+            # rank-consistent decision: if any DP rank failed to capture,
+            # all ranks run eager so collective sequences stay aligned
+            from ..parallel.dist import agree_all
+            if not agree_all(self.use_graph):
+                self.use_graph = False
+                self._graph = None
         if self.use_graph:
             self._graph.replay()
         else:

@@ -99,3 +99,19 @@ def allreduce_scalar(x: float, op: str = "max") -> float:
                      device="cuda" if dist.get_backend() == "nccl" else "cpu")
     dist.all_reduce(t, op=dist.ReduceOp.MAX if op == "max" else dist.ReduceOp.SUM)
     return float(t.item())
+
+
+def agree_all(ok: bool) -> bool:
+    """Collective AND across ranks (True if every rank passed True).
+
+    Used to make per-rank runtime decisions (e.g. hipGraph capture
+    success) rank-consistent: if any rank failed, all ranks must take
+    the fallback path so every iteration issues the same collective
+    sequence on every rank. No-op outside distributed runs.
+    """
+    if not dist.is_initialized():
+        return ok
+    t = torch.tensor([1.0 if ok else 0.0], dtype=torch.float64,
+                     device="cuda" if dist.get_backend() == "nccl" else "cpu")
+    dist.all_reduce(t, op=dist.ReduceOp.MIN)
+    return bool(t.item() > 0.5)

@@ -93,3 +93,24 @@ def test_distributed_ppo_training_gloo():
     assert len(res) == 2
     for rank, loss in res:
         assert np.isfinite(loss)
+
+
+def _worker_agree(rank, world, port, q):
+    import os
+    os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+                      RANK=str(rank), WORLD_SIZE=str(world))
+    import torch.distributed as d
+    d.init_process_group("gloo", rank=rank, world_size=world)
+    from rlgpuschedule_amd.parallel.dist import agree_all
+    # mixed votes -> everyone must get False; unanimous True -> True
+    r1 = agree_all(rank == 0)
+    r2 = agree_all(True)
+    q.put((rank, r1, r2))
+    d.destroy_process_group()
+
+
+def test_agree_all_gloo():
+    res = _spawn(_worker_agree)
+    for rank, r1, r2 in res:
+        assert r1 is False    # one rank voted False -> all False
+        assert r2 is True
