@@ -43,6 +43,13 @@ __device__ __forceinline__ int wave_sum_i32(int v) {
   return v;
 }
 
+__device__ __forceinline__ int wave_max_i32(int v) {
+  #pragma unroll
+  for (int off = 32; off > 0; off >>= 1)
+    v = max(v, __shfl_xor(v, off, RLS_WAVE));
+  return v;
+}
+
 // float32 math mirrored by cpu_ref._lognorm
 __device__ __forceinline__ float lognorm(int64_t x_us) {
   return log1pf((float)x_us / 1e6f) / 10.0f;
@@ -374,6 +381,36 @@ __device__ void write_obs_mask(const EnvView& v, const Sc& s, int lane) {
       v.obs[bix + 4] = fits ? 1.0f : 0.0f;
     }
     v.mask[lane] = (uint8_t)(valid && fits);
+  }
+  // obs v2 extras (ENV_SPEC v4): written after the base layout so the
+  // default obs_version=1 path is bit-identical to the original kernel
+  if (v.p->obs_version >= 2) {
+    const int EB = 4 + 5 * K, G = (int)v.p->G;
+    int cnt[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+    int mxf = 0;
+    for (int n = lane; n < N; n += RLS_WAVE) {
+      int f = v.node_free[n];
+      int idx = (f * 8) / G;
+      if (idx > 7) idx = 7;
+      mxf = max(mxf, f);
+      #pragma unroll
+      for (int b = 0; b < 8; ++b) cnt[b] += (idx == b) ? 1 : 0;
+    }
+    mxf = wave_max_i32(mxf);
+    #pragma unroll
+    for (int b = 0; b < 8; ++b) {
+      int tot = wave_sum_i32(cnt[b]);
+      if (lane == 0) v.obs[EB + b] = (float)tot / (float)N;
+    }
+    if (lane == 0) {
+      int over = s.pend_count - K;
+      v.obs[EB + 8] = (over > 0 ? (float)over : 0.0f) / (float)v.p->P;
+      v.obs[EB + 9] = (float)mxf / (float)G;
+      int64_t span = v.job_hi - v.job_lo;
+      if (span < 1) span = 1;
+      v.obs[EB + 10] = (float)(s.cursor - v.job_lo) / (float)span;
+      v.obs[EB + 11] = (float)s.decisions / (float)v.p->horizon;
+    }
   }
   if (lane == (int)(K & 63)) v.mask[K] = 1;
   if (v.p->gandiva && lane == 0) {

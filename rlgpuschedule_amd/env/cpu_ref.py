@@ -289,6 +289,19 @@ class ClusterEnvRef:
             obs[b + 3] = _lognorm(self._pend_work_us(k))
             obs[b + 4] = np.float32(
                 1.0 if self.first_fit(g_fit) is not None else 0.0)
+        if s.obs_version >= 2:
+            e = 4 + 5 * s.queue_visible
+            gpn = s.gpus_per_node
+            for n in range(s.num_nodes):
+                idx = min(7, (int(self.node_free[n]) * 8) // gpn)
+                obs[e + idx] += np.float32(1.0) / np.float32(s.num_nodes)
+            obs[e + 8] = (np.float32(max(0, self.pend_count - s.queue_visible))
+                          / np.float32(s.queue_cap))
+            obs[e + 9] = (np.float32(int(self.node_free.max()))
+                          / np.float32(gpn))
+            span = max(1, self.job_hi - self.job_lo)
+            obs[e + 10] = np.float32(self.cursor - self.job_lo) / np.float32(span)
+            obs[e + 11] = np.float32(self.decisions) / np.float32(s.horizon)
         return obs
 
     # integer-state snapshot for differential tests against the HIP env

@@ -82,6 +82,7 @@ class HipVectorEnv:
             1 if spec.gandiva else 0,
             1 if spec.elastic else 0,
             int(debug_skip),
+            int(spec.obs_version),
         ], dtype=np.int64)
         self.binding = EnvStepBinding(
             params, {k: v.data_ptr() for k, v in self.b.items()})

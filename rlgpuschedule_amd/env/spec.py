@@ -22,6 +22,7 @@ class EnvSpec:
     horizon: int = 50_000
     gandiva: bool = False            # adds MIGRATE/TIMESLICE actions (v2)
     elastic: bool = False            # elastic multi-GPU jobs (v3)
+    obs_version: int = 1             # 2 adds 12 global features (v4)
 
     def __post_init__(self):
         if self.run_cap == 0:
@@ -51,9 +52,15 @@ class EnvSpec:
         assert self.gandiva
         return self.queue_visible + 2
 
+    # obs v2 extras: 8-bucket node free-fraction histogram, pending
+    # overflow beyond the visible window, max single-node free fraction,
+    # trace-consumption progress, episode progress (docs/ENV_SPEC.md v4)
+    OBS_V2_EXTRA = 12
+
     @property
     def obs_dim(self) -> int:
-        return 4 + 5 * self.queue_visible
+        return (4 + 5 * self.queue_visible
+                + (self.OBS_V2_EXTRA if self.obs_version >= 2 else 0))
 
     @property
     def max_job_gpus(self) -> int:

@@ -298,3 +298,35 @@ def test_invariants_fuzz():
             if done:
                 env.reset()
         env.validate_invariants()
+
+
+def test_obs_v2_extras():
+    """obs_version=2: histogram sums to 1, extras well-formed, and the
+    default obs_version=1 layout is a strict prefix (bit-identical)."""
+    import numpy as np
+
+    from rlgpuschedule_amd.env.cpu_ref import ClusterEnvRef
+    from rlgpuschedule_amd.env.spec import EnvSpec
+    from rlgpuschedule_amd.traces.synthetic import philly_like
+    s1 = EnvSpec(num_nodes=4, gpus_per_node=8, horizon=500)
+    s2 = EnvSpec(num_nodes=4, gpus_per_node=8, horizon=500, obs_version=2)
+    assert s2.obs_dim == s1.obs_dim + EnvSpec.OBS_V2_EXTRA
+    tr = philly_like(24, seed=5, max_job_gpus=s1.max_job_gpus,
+                     mean_interarrival_s=20.0, median_duration_s=120.0)
+    e1, e2 = ClusterEnvRef(s1, tr), ClusterEnvRef(s2, tr)
+    o1, o2 = e1.reset(), e2.reset()
+    rng = np.random.default_rng(3)
+    for t in range(120):
+        m1 = e1.action_mask()
+        a = int(rng.choice(np.flatnonzero(m1)))
+        o1 = e1.step(a)[0]
+        o2, _, d2, _ = e2.step(a)
+        base = s1.obs_dim
+        np.testing.assert_array_equal(o1, o2[:base])   # prefix identical
+        ex = o2[base:]
+        assert abs(ex[:8].sum() - 1.0) < 1e-5          # histogram
+        assert ((ex >= 0) & (ex <= 1.0 + 1e-6)).all()
+        # cross-check: max free fraction consistent with node_free
+        assert abs(ex[9] - e2.node_free.max() / s2.gpus_per_node) < 1e-6
+        if d2:
+            break

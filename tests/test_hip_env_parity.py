@@ -150,3 +150,20 @@ def test_state_invariants_under_load():
         env.step(acts)
         if i % 100 == 99:
             env.validate_state()
+
+
+@pytest.mark.gpu
+def test_parity_obs_v2():
+    """obs_version=2 extras (node histogram, overflow, progress) parity."""
+    spec = EnvSpec(num_nodes=12, gpus_per_node=8, queue_visible=10,
+                   queue_cap=64, horizon=2000, obs_version=2)
+    _run_parity(spec, jobs_per_env=24, steps=300,
+                mean_interarrival_s=20.0, median_duration_s=120.0)
+
+
+@pytest.mark.gpu
+def test_parity_obs_v2_gandiva_elastic():
+    spec = EnvSpec(num_nodes=6, gpus_per_node=4, queue_visible=8,
+                   horizon=1500, gandiva=True, elastic=True, obs_version=2)
+    _run_parity(spec, jobs_per_env=20, steps=250,
+                mean_interarrival_s=25.0, median_duration_s=150.0)
