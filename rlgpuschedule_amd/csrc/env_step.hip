@@ -523,6 +523,7 @@ void unpack(const int64_t* params, const uint64_t* bufs, RlsParams& p, RlsBufs& 
   p.obs_dim = params[8]; p.idle_tick_us = params[9]; p.time_scale_us = params[10];
   p.horizon = params[11]; p.gandiva = params[12]; p.elastic = params[13];
   p.debug_skip = params[14];
+  p.obs_version = params[15];
   uint64_t* dst = reinterpret_cast<uint64_t*>(&b);
   for (int i = 0; i < RLS_NUM_BUFS; ++i) dst[i] = bufs[i];
 }
