@@ -27,6 +27,8 @@ def main(argv=None):
     tr.add_argument("--resume", default=None, help="checkpoint path")
     tr.add_argument("--actor-dtype", choices=["fp32", "bf16"], default=None)
     tr.add_argument("--update-dtype", choices=["fp32", "bf16"], default=None)
+    tr.add_argument("--obs-version", type=int, choices=[1, 2], default=None,
+                    help="2 = richer observations (docs/ENV_SPEC.md v4)")
     ev = sub.add_parser("eval", help="evaluate a checkpoint or baseline")
     ev.add_argument("--config", choices=sorted(CONFIGS), default="c2_ppo_mlp")
     ev.add_argument("--ckpt", default=None, help="policy checkpoint (.pt/.safetensors)")
@@ -120,6 +122,9 @@ def main(argv=None):
         over["actor_dtype"] = args.actor_dtype
     if args.update_dtype is not None:
         over["update_dtype"] = args.update_dtype
+    if args.obs_version is not None:
+        over["spec"] = dataclasses.replace(cfg.spec,
+                                           obs_version=args.obs_version)
     if over:
         cfg = dataclasses.replace(cfg, **over)
     out = train(cfg, iterations=args.iterations, out_dir=args.out,
