@@ -224,6 +224,78 @@ class ClusterEnvRef:
         self._release(slot)
         return True
 
+    def _shrink(self) -> bool:
+        """ENV_SPEC v5 SHRINK: halve the widest running elastic job
+        (largest allocated g with g//2 >= min_gpus; tie -> lowest slot).
+        Frees GPUs for the queue at the cost of slower progress. False
+        (degrade to NOOP) if no candidate or the re-placement fails."""
+        best_g, slot = -1, -1
+        for r in range(self.spec.run_cap):
+            if not self.run_valid[r]:
+                continue
+            g = int(self.run_gpus[r])
+            j = int(self.run_job[r])
+            if g >= 2 and g // 2 >= int(self.trace.min_gpus[j]):
+                if g > best_g:
+                    best_g, slot = g, r
+        if slot < 0:
+            return False
+        g_old = best_g
+        g_new = g_old // 2
+        rem = max(1, int(self.run_finish[slot]) - self.clock) * g_old
+        pairs_old = self.run_alloc[slot]
+        for n, c in pairs_old:
+            self.node_free[n] += c
+        pairs_new = self.first_fit(g_new)
+        if pairs_new is None:   # span-constrained fit can fail: restore
+            for n, c in pairs_old:
+                self.node_free[n] -= c
+            return False
+        for n, c in pairs_new:
+            self.node_free[n] -= c
+        self.run_alloc[slot] = pairs_new
+        self.run_gpus[slot] = g_new
+        self.run_finish[slot] = self.clock + (-(-rem // g_new))
+        return True
+
+    def _expand(self) -> bool:
+        """ENV_SPEC v5 EXPAND: grow the most-shrunk running job (largest
+        demand deficit gpus - g; tie -> lowest slot) toward its demand,
+        doubling at most (g_target = min(gpus, 2g)), taking the largest
+        g' in (g, g_target] that fits. False if no candidate / no fit."""
+        best_d, slot = 0, -1
+        for r in range(self.spec.run_cap):
+            if not self.run_valid[r]:
+                continue
+            d = int(self.trace.gpus[int(self.run_job[r])]) - int(self.run_gpus[r])
+            if d > best_d:
+                best_d, slot = d, r
+        if slot < 0:
+            return False
+        g_old = int(self.run_gpus[slot])
+        j = int(self.run_job[slot])
+        g_target = min(int(self.trace.gpus[j]), 2 * g_old)
+        rem = max(1, int(self.run_finish[slot]) - self.clock) * g_old
+        pairs_old = self.run_alloc[slot]
+        for n, c in pairs_old:
+            self.node_free[n] += c
+        g_new, pairs_new = -1, None
+        for gt in range(g_target, g_old, -1):
+            pairs_new = self.first_fit(gt)
+            if pairs_new is not None:
+                g_new = gt
+                break
+        if g_new < 0:
+            for n, c in pairs_old:
+                self.node_free[n] -= c
+            return False
+        for n, c in pairs_new:
+            self.node_free[n] -= c
+        self.run_alloc[slot] = pairs_new
+        self.run_gpus[slot] = g_new
+        self.run_finish[slot] = self.clock + (-(-rem // g_new))
+        return True
+
     def step(self, action: int) -> tuple[np.ndarray, np.float32, bool, dict]:
         s = self.spec
         reward = np.float32(0.0)
@@ -236,6 +308,10 @@ class ClusterEnvRef:
             did = self._migrate()
         elif s.gandiva and a == s.timeslice_action:
             did = self._timeslice()
+        elif s.elastic_resize and a == s.shrink_action:
+            did = self._shrink()
+        elif s.elastic_resize and a == s.expand_action:
+            did = self._expand()
         if not did:
             reward = self._advance()
         self.decisions += 1
@@ -267,6 +343,19 @@ class ClusterEnvRef:
             m[s.migrate_action] = self.run_count > 0
             m[s.timeslice_action] = (self.run_count > 0
                                      and self.pend_count < s.queue_cap)
+        if s.elastic_resize:
+            can_s = can_e = False
+            for r in range(s.run_cap):
+                if not self.run_valid[r]:
+                    continue
+                g = int(self.run_gpus[r])
+                j = int(self.run_job[r])
+                if g >= 2 and g // 2 >= int(self.trace.min_gpus[j]):
+                    can_s = True
+                if g < int(self.trace.gpus[j]):
+                    can_e = True
+            m[s.shrink_action] = can_s
+            m[s.expand_action] = can_e
         return m
 
     def observe(self) -> np.ndarray:

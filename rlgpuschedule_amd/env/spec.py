@@ -23,12 +23,15 @@ class EnvSpec:
     gandiva: bool = False            # adds MIGRATE/TIMESLICE actions (v2)
     elastic: bool = False            # elastic multi-GPU jobs (v3)
     obs_version: int = 1             # 2 adds 12 global features (v4)
+    elastic_resize: bool = False     # SHRINK/EXPAND running jobs (v5)
 
     def __post_init__(self):
         if self.run_cap == 0:
             object.__setattr__(self, "run_cap", min(self.total_gpus, 512))
         assert self.num_nodes >= 1 and self.gpus_per_node >= 1
         assert self.queue_visible <= self.queue_cap
+        assert not self.elastic_resize or self.elastic, \
+            "elastic_resize requires elastic jobs (min_gpus < gpus)" 
 
     @property
     def total_gpus(self) -> int:
@@ -36,7 +39,8 @@ class EnvSpec:
 
     @property
     def num_actions(self) -> int:
-        return self.queue_visible + 1 + (2 if self.gandiva else 0)
+        return (self.queue_visible + 1 + (2 if self.gandiva else 0)
+                + (2 if self.elastic_resize else 0))
 
     @property
     def noop_action(self) -> int:
@@ -51,6 +55,15 @@ class EnvSpec:
     def timeslice_action(self) -> int:
         assert self.gandiva
         return self.queue_visible + 2
+
+    @property
+    def shrink_action(self) -> int:
+        assert self.elastic_resize
+        return self.queue_visible + 1 + (2 if self.gandiva else 0)
+
+    @property
+    def expand_action(self) -> int:
+        return self.shrink_action + 1
 
     # obs v2 extras: 8-bucket node free-fraction histogram, pending
     # overflow beyond the visible window, max single-node free fraction,

@@ -330,3 +330,67 @@ def test_obs_v2_extras():
         assert abs(ex[9] - e2.node_free.max() / s2.gpus_per_node) < 1e-6
         if d2:
             break
+
+
+def test_elastic_resize_shrink_expand():
+    """ENV_SPEC v5: SHRINK halves the widest elastic job (work conserved in
+    GPU-us), EXPAND grows it back toward demand; masks reflect candidates."""
+    import numpy as np
+
+    from rlgpuschedule_amd.env.cpu_ref import ClusterEnvRef
+    from rlgpuschedule_amd.env.spec import EnvSpec
+    from rlgpuschedule_amd.traces.synthetic import JobTrace
+    spec = EnvSpec(num_nodes=2, gpus_per_node=8, queue_visible=4,
+                   horizon=1000, elastic=True, elastic_resize=True)
+    tr = JobTrace(arrival_us=np.array([0, 0], np.int64),
+                  gpus=np.array([8, 4], np.int32),
+                  work_us=np.array([8_000_000, 4_000_000], np.int64),
+                  min_gpus=np.array([2, 1], np.int32))
+    env = ClusterEnvRef(spec, tr)
+    env.reset()
+    assert env.step(0)[0] is not None        # place job0 at g=8
+    assert env.step(0)[0] is not None        # place job1 at g=4
+    g0 = int(env.run_gpus[0])
+    assert g0 == 8 and int(env.run_gpus[1]) == 4
+    f0 = int(env.run_finish[0])
+    m = env.action_mask()
+    assert m[spec.shrink_action]             # both jobs shrinkable
+    assert not m[spec.expand_action]         # nobody below demand
+    # SHRINK picks the widest (job0, g=8 -> 4); remaining work conserved
+    rem_before = (f0 - env.clock) * 8
+    assert env.step(spec.shrink_action)
+    assert int(env.run_gpus[0]) == 4
+    assert int(env.run_finish[0]) - env.clock == -(-rem_before // 4)
+    env.validate_invariants()
+    m = env.action_mask()
+    assert m[spec.expand_action]             # job0 now below demand
+    # EXPAND doubles it back (4 free GPUs exist): 4 -> 8
+    assert env.step(spec.expand_action)
+    assert int(env.run_gpus[0]) == 8
+    env.validate_invariants()
+    assert not env.action_mask()[spec.expand_action]
+
+
+def test_elastic_resize_expand_partial_fit():
+    """EXPAND takes the largest g' in (g, 2g] that fits, not all-or-nothing."""
+    import numpy as np
+
+    from rlgpuschedule_amd.env.cpu_ref import ClusterEnvRef
+    from rlgpuschedule_amd.env.spec import EnvSpec
+    from rlgpuschedule_amd.traces.synthetic import JobTrace
+    spec = EnvSpec(num_nodes=1, gpus_per_node=8, queue_visible=4,
+                   horizon=1000, elastic=True, elastic_resize=True)
+    tr = JobTrace(arrival_us=np.array([0, 0], np.int64),
+                  gpus=np.array([8, 3], np.int32),
+                  work_us=np.array([8_000_000, 9_000_000], np.int64),
+                  min_gpus=np.array([2, 3], np.int32))
+    env = ClusterEnvRef(spec, tr)
+    env.reset()
+    env.step(0)                               # job0 g=8 (full node)
+    env.step(spec.shrink_action)              # -> g=4, 4 free
+    env.step(0)                               # job1 g=3, 1 free
+    assert int(env.run_gpus[0]) == 4 and int(env.run_gpus[1]) == 3
+    # expand job0: target min(8, 8)=8, only 5 total available -> g'=5
+    assert env.step(spec.expand_action)
+    assert int(env.run_gpus[0]) == 5
+    env.validate_invariants()

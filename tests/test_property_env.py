@@ -14,9 +14,11 @@ from rlgpuschedule_amd.env.spec import EnvSpec
 from rlgpuschedule_amd.traces.synthetic import philly_like
 
 
-def _mk_env(nodes, gpn, gandiva, elastic, jobs, seed, obs_version=1):
+def _mk_env(nodes, gpn, gandiva, elastic, jobs, seed, obs_version=1,
+            resize=False):
     spec = EnvSpec(num_nodes=nodes, gpus_per_node=gpn, horizon=80_000,
-                   gandiva=gandiva, elastic=elastic, obs_version=obs_version)
+                   gandiva=gandiva, elastic=elastic, obs_version=obs_version,
+                   elastic_resize=resize and elastic)
     tr = philly_like(jobs, seed=seed, max_job_gpus=spec.max_job_gpus,
                      mean_interarrival_s=40.0, median_duration_s=300.0,
                      elastic=elastic)
@@ -27,11 +29,12 @@ def _mk_env(nodes, gpn, gandiva, elastic, jobs, seed, obs_version=1):
 @given(nodes=st.integers(1, 6), gpn=st.sampled_from([2, 4, 8]),
        gandiva=st.booleans(), elastic=st.booleans(),
        jobs=st.integers(4, 48), seed=st.integers(0, 10_000),
-       aseed=st.integers(0, 10_000), obs_version=st.sampled_from([1, 2]))
+       aseed=st.integers(0, 10_000), obs_version=st.sampled_from([1, 2]),
+       resize=st.booleans())
 def test_invariants_hold_under_random_masked_actions(
-        nodes, gpn, gandiva, elastic, jobs, seed, aseed, obs_version):
+        nodes, gpn, gandiva, elastic, jobs, seed, aseed, obs_version, resize):
     spec, env = _mk_env(nodes, gpn, gandiva, elastic, jobs, seed,
-                        obs_version)
+                        obs_version, resize)
     env.reset()
     rng = np.random.default_rng(aseed)
     for t in range(200):
