@@ -28,6 +28,7 @@ struct RlsParams {
   int64_t debug_skip; // ablation bitmask (bench only): 1=slot obs/fits,
                       // 2=whole obs/mask phase, 4=obs[3] next_event
   int64_t obs_version; // 1 = base layout; 2 adds 12 global features (v4)
+  int64_t elastic_resize; // 0/1: SHRINK/EXPAND running jobs (v5)
 };
 
 // Device buffers; all row-major with env as the leading dim.

@@ -347,6 +347,91 @@ __device__ bool do_timeslice(const EnvView& v, Sc& s) {
   return true;
 }
 
+// ENV_SPEC v5 SHRINK: halve the widest running elastic job (largest g
+// with g/2 >= min_gpus; tie -> lowest slot). Redundant-scalar.
+__device__ bool do_shrink(const EnvView& v, Sc& s) {
+  const int R = (int)v.p->R, S = (int)v.p->S;
+  int best_g = -1, slot = -1;
+  for (int r = 0; r < R; ++r) {
+    if (!v.run_valid[r]) continue;
+    int g = v.run_gpus[r];
+    int32_t j = v.run_job[r];
+    if (g >= 2 && g / 2 >= v.trace_ming[j] && g > best_g) {
+      best_g = g; slot = r;
+    }
+  }
+  if (slot < 0) return false;
+  int g_old = best_g, g_new = g_old / 2;
+  int64_t rem0 = s.clock < v.run_finish[slot]
+                 ? v.run_finish[slot] - s.clock : 1;
+  int64_t rem = rem0 * g_old;
+  int old_pk[RLS_MAX_SPAN];
+  for (int i = 0; i < S; ++i) {
+    old_pk[i] = v.run_alloc[slot * S + i];
+    int cnt = old_pk[i] & 0xFFFF;
+    if (cnt > 0) v.node_free[old_pk[i] >> 16] += cnt;
+  }
+  int pn[RLS_MAX_SPAN], pc_[RLS_MAX_SPAN];
+  int span = first_fit(v, g_new, pn, pc_);
+  if (span < 0) {  // span-constrained fit can fail: restore, NOOP
+    for (int i = 0; i < S; ++i) {
+      int cnt = old_pk[i] & 0xFFFF;
+      if (cnt > 0) v.node_free[old_pk[i] >> 16] -= cnt;
+    }
+    return false;
+  }
+  for (int i = 0; i < span; ++i) v.node_free[pn[i]] -= pc_[i];
+  for (int i = 0; i < S; ++i)
+    v.run_alloc[slot * S + i] = i < span ? ((pn[i] << 16) | pc_[i]) : 0;
+  v.run_gpus[slot] = g_new;
+  v.run_finish[slot] = s.clock + (rem + g_new - 1) / g_new;
+  return true;
+}
+
+// ENV_SPEC v5 EXPAND: grow the most-shrunk running job (largest demand
+// deficit; tie -> lowest slot) to the largest g' in (g, min(demand, 2g)].
+__device__ bool do_expand(const EnvView& v, Sc& s) {
+  const int R = (int)v.p->R, S = (int)v.p->S;
+  int best_d = 0, slot = -1;
+  for (int r = 0; r < R; ++r) {
+    if (!v.run_valid[r]) continue;
+    int d = v.trace_gpus[v.run_job[r]] - v.run_gpus[r];
+    if (d > best_d) { best_d = d; slot = r; }
+  }
+  if (slot < 0) return false;
+  int g_old = v.run_gpus[slot];
+  int32_t j = v.run_job[slot];
+  int g_target = min((int)v.trace_gpus[j], 2 * g_old);
+  int64_t rem0 = s.clock < v.run_finish[slot]
+                 ? v.run_finish[slot] - s.clock : 1;
+  int64_t rem = rem0 * g_old;
+  int old_pk[RLS_MAX_SPAN];
+  for (int i = 0; i < S; ++i) {
+    old_pk[i] = v.run_alloc[slot * S + i];
+    int cnt = old_pk[i] & 0xFFFF;
+    if (cnt > 0) v.node_free[old_pk[i] >> 16] += cnt;
+  }
+  int pn[RLS_MAX_SPAN], pc_[RLS_MAX_SPAN];
+  int g_new = -1, span = -1;
+  for (int gt = g_target; gt > g_old; --gt) {
+    span = first_fit(v, gt, pn, pc_);
+    if (span >= 0) { g_new = gt; break; }
+  }
+  if (g_new < 0) {
+    for (int i = 0; i < S; ++i) {
+      int cnt = old_pk[i] & 0xFFFF;
+      if (cnt > 0) v.node_free[old_pk[i] >> 16] -= cnt;
+    }
+    return false;
+  }
+  for (int i = 0; i < span; ++i) v.node_free[pn[i]] -= pc_[i];
+  for (int i = 0; i < S; ++i)
+    v.run_alloc[slot * S + i] = i < span ? ((pn[i] << 16) | pc_[i]) : 0;
+  v.run_gpus[slot] = g_new;
+  v.run_finish[slot] = s.clock + (rem + g_new - 1) / g_new;
+  return true;
+}
+
 // Observation + action mask (ENV_SPEC layout). Wave-parallel over slots.
 __device__ void write_obs_mask(const EnvView& v, const Sc& s, int lane) {
   const int K = (int)v.p->K, N = (int)v.p->N, A = (int)v.p->A;
@@ -416,6 +501,23 @@ __device__ void write_obs_mask(const EnvView& v, const Sc& s, int lane) {
   if (v.p->gandiva && lane == 0) {
     v.mask[K + 1] = (uint8_t)(s.run_count > 0);
     v.mask[K + 2] = (uint8_t)(s.run_count > 0 && s.pend_count < (int)v.p->P);
+  }
+  if (v.p->elastic_resize) {   // v5: wave-parallel candidate scan
+    const int R = (int)v.p->R;
+    bool can_s = false, can_e = false;
+    for (int r = lane; r < R; r += RLS_WAVE) {
+      if (!v.run_valid[r]) continue;
+      int g = v.run_gpus[r];
+      int32_t j = v.run_job[r];
+      can_s |= (g >= 2 && g / 2 >= v.trace_ming[j]);
+      can_e |= (g < v.trace_gpus[j]);
+    }
+    uint64_t bs = __ballot(can_s), be = __ballot(can_e);
+    if (lane == 0) {
+      int base = K + 1 + (v.p->gandiva ? 2 : 0);
+      v.mask[base] = (uint8_t)(bs != 0);
+      v.mask[base + 1] = (uint8_t)(be != 0);
+    }
   }
 }
 
@@ -496,6 +598,10 @@ __global__ __launch_bounds__(RLS_BLOCK) void env_step_kernel(RlsParams p, RlsBuf
     did = do_migrate(v, s);
   } else if (p.gandiva && a == p.K + 2) {
     did = do_timeslice(v, s);
+  } else if (p.elastic_resize) {
+    int64_t base = p.K + 1 + (p.gandiva ? 2 : 0);
+    if (a == base) did = do_shrink(v, s);
+    else if (a == base + 1) did = do_expand(v, s);
   }
   if (!did) reward = advance(v, s, lane);
   s.decisions += 1;
@@ -524,6 +630,7 @@ void unpack(const int64_t* params, const uint64_t* bufs, RlsParams& p, RlsBufs& 
   p.horizon = params[11]; p.gandiva = params[12]; p.elastic = params[13];
   p.debug_skip = params[14];
   p.obs_version = params[15];
+  p.elastic_resize = params[16];
   uint64_t* dst = reinterpret_cast<uint64_t*>(&b);
   for (int i = 0; i < RLS_NUM_BUFS; ++i) dst[i] = bufs[i];
 }

@@ -83,6 +83,7 @@ class HipVectorEnv:
             1 if spec.elastic else 0,
             int(debug_skip),
             int(spec.obs_version),
+            1 if spec.elastic_resize else 0,
         ], dtype=np.int64)
         self.binding = EnvStepBinding(
             params, {k: v.data_ptr() for k, v in self.b.items()})

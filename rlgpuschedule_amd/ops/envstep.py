@@ -44,7 +44,7 @@ def load() -> ctypes.CDLL:
 
 class EnvStepBinding:
     def __init__(self, params: np.ndarray, buf_ptrs: dict[str, int]):
-        assert params.dtype == np.int64 and len(params) == 16
+        assert params.dtype == np.int64 and len(params) == 17
         self.lib = load()
         self.params = np.ascontiguousarray(params)
         self._ptrs = dict(buf_ptrs)

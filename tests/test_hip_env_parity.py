@@ -167,3 +167,22 @@ def test_parity_obs_v2_gandiva_elastic():
                    horizon=1500, gandiva=True, elastic=True, obs_version=2)
     _run_parity(spec, jobs_per_env=20, steps=250,
                 mean_interarrival_s=25.0, median_duration_s=150.0)
+
+
+@pytest.mark.gpu
+def test_parity_elastic_resize():
+    """ENV_SPEC v5 SHRINK/EXPAND parity under random masked actions."""
+    spec = EnvSpec(num_nodes=6, gpus_per_node=8, queue_visible=8,
+                   horizon=2500, elastic=True, elastic_resize=True)
+    _run_parity(spec, jobs_per_env=24, steps=350,
+                mean_interarrival_s=15.0, median_duration_s=200.0)
+
+
+@pytest.mark.gpu
+def test_parity_everything_on():
+    """All feature gates at once: gandiva + elastic + resize + obs v2."""
+    spec = EnvSpec(num_nodes=8, gpus_per_node=4, queue_visible=10,
+                   horizon=2000, gandiva=True, elastic=True,
+                   elastic_resize=True, obs_version=2)
+    _run_parity(spec, jobs_per_env=20, steps=300,
+                mean_interarrival_s=20.0, median_duration_s=150.0)
