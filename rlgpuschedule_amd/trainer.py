@@ -53,7 +53,8 @@ CONFIGS = {
         algo="ppo", policy="gru"),
     # Gandiva actions + elastic jobs, A2C
     "c4_gandiva_a2c": TrainConfig(
-        name="c4_gandiva_a2c", spec=cluster64(gandiva=True, elastic=True),
+        name="c4_gandiva_a2c", spec=cluster64(gandiva=True, elastic=True,
+                       elastic_resize=True),
         num_envs=1024, rollout_T=64, algo="a2c", policy="mlp",
         trace_kw={"elastic": True, "fail_frac": 0.1}),
     # 1024-GPU Poisson trace, IMPALA-style async actor/learner
