@@ -561,6 +561,7 @@ struct AdamArgs {
   int64_t R[ADAM_MAX_SEG], C[ADAM_MAX_SEG];
   int64_t c0[ADAM_MAX_SEG], c1[ADAM_MAX_SEG];
   int64_t cum[ADAM_MAX_SEG + 1];   // cumulative element counts
+  int64_t rmsprop;                 // 0 = Adam, 1 = RMSprop (alpha=beta2)
   float* param[ADAM_MAX_SEG];
   int64_t transposed[ADAM_MAX_SEG];
 };
@@ -591,12 +592,19 @@ __global__ __launch_bounds__(256) void adam_step_kernel(AdamArgs a) {
   const int64_t c = a.c0[seg] + cc;
   float g = a.gflat[a.goff[seg] + r * a.C[seg] + c] * a.coef[0];
   const int64_t mi = a.moff[seg] + e;
-  float m = a.beta1 * a.m[mi] + (1.0f - a.beta1) * g;
-  float v = a.beta2 * a.v[mi] + (1.0f - a.beta2) * g * g;
-  a.m[mi] = m;
-  a.v[mi] = v;
-  const float denom = __fsqrt_rn(v) / __fsqrt_rn(a.bc[1]) + a.eps;
-  const float upd = (a.lr / a.bc[0]) * m / denom;
+  float upd;
+  if (a.rmsprop) {   // torch.optim.RMSprop (momentum 0, uncentered)
+    float v = a.beta2 * a.v[mi] + (1.0f - a.beta2) * g * g;
+    a.v[mi] = v;
+    upd = a.lr * g / (__fsqrt_rn(v) + a.eps);
+  } else {
+    float m = a.beta1 * a.m[mi] + (1.0f - a.beta1) * g;
+    float v = a.beta2 * a.v[mi] + (1.0f - a.beta2) * g * g;
+    a.m[mi] = m;
+    a.v[mi] = v;
+    const float denom = __fsqrt_rn(v) / __fsqrt_rn(a.bc[1]) + a.eps;
+    upd = (a.lr / a.bc[0]) * m / denom;
+  }
   float* p = a.param[seg];
   // param layouts: transposed weights [outW, R]; sliced/non-transposed
   // segments are compact [R, W] (W = c1-c0; biases have R == 1)
@@ -644,6 +652,7 @@ extern "C" int rlsched_adam_step(const int64_t* ip, const double* fp,
     a.param[s] = (float*)bufs[5 + s];
     a.cum[s + 1] = a.cum[s] + a.R[s] * (a.c1[s] - a.c0[s]);
   }
+  a.rmsprop = ip[1 + 7 * a.nseg];
   dim3 grid((unsigned)((a.cum[a.nseg] + 255) / 256));
   hipLaunchKernelGGL(adam_step_kernel, grid, dim3(256), 0,
                      (hipStream_t)stream, a);

@@ -136,7 +136,9 @@ class HipAdam:
     """
 
     def __init__(self, gflat, segments, lr=3e-4, betas=(0.9, 0.999),
-                 eps=1e-5, max_norm=0.5, grad_reduce=None):
+                 eps=1e-5, max_norm=0.5, grad_reduce=None, algo="adam"):
+        assert algo in ("adam", "rmsprop")  # rmsprop: betas[1] is alpha
+        self.algo = algo
         self.gflat = gflat
         self.lr, self.betas, self.eps = lr, betas, eps
         self.max_norm = max_norm
@@ -154,6 +156,7 @@ class HipAdam:
             ip += [s["goff"], moff, s["R"], s["C"], s["c0"], s["c1"]]
             moff += s["R"] * (s["c1"] - s["c0"])
         ip += [1 if s["transposed"] else 0 for s in segments]
+        ip += [1 if algo == "rmsprop" else 0]
         self._ip = np.asarray(ip, np.int64)
         self._bump_ip = np.zeros(1, np.int64)
         self._bump_fp = np.asarray(betas, np.float64)

@@ -135,6 +135,10 @@ def train(cfg: TrainConfig, iterations: int | None = None,
         from .agents.ppo_gru_hip import HipGRUPPOLearner
         learner = HipGRUPPOLearner(policy, cfg.ppo, grad_sync,
                                    dtype=cfg.update_dtype)
+    elif (on_gpu_fast and cfg.algo == "a2c" and not policy.is_recurrent):
+        from .agents.a2c_hip import HipA2CLearner
+        learner = HipA2CLearner(policy, cfg.a2c, grad_sync,
+                                dtype=cfg.update_dtype)
     else:
         learner = make_learner(cfg, policy, grad_sync)
     start_it = 0

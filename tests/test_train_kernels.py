@@ -281,3 +281,45 @@ def test_hip_ppo_kl_penalty_matches_reference():
     for (na, pa), (nb, pb) in zip(pol_a.named_parameters(),
                                   pol_b.named_parameters()):
         torch.testing.assert_close(pa, pb, atol=2e-4, rtol=1e-3)
+
+
+@pytest.mark.gpu
+def test_hip_a2c_learner_matches_reference():
+    """HipA2CLearner (fused head + RMSprop-mode step) vs torch A2CLearner."""
+    import copy
+
+    from helpers import make_buf as _make_buf
+    from rlgpuschedule_amd.agents.a2c_hip import HipA2CLearner
+    from rlgpuschedule_amd.agents.ppo import A2CConfig, A2CLearner
+    from rlgpuschedule_amd.models.policy import MLPPolicy
+    torch.manual_seed(5)
+    spec = cluster64()
+    T, E = 16, 64
+    pol_a = MLPPolicy(spec.obs_dim, spec.num_actions, 256).cuda()
+    pol_b = copy.deepcopy(pol_a)
+    cfg = A2CConfig(norm_reward=False)
+    ref = A2CLearner(pol_a, cfg)
+    hip = HipA2CLearner(pol_b, cfg, use_graph=False)
+    buf = _make_buf(T, E, spec, "cuda", seed=14)
+    s_ref = ref.update(buf)
+    s_hip = hip.update(buf)
+    for k in ("pi_loss", "v_loss", "entropy"):
+        assert abs(s_ref[k] - s_hip[k]) < 5e-3, (k, s_ref[k], s_hip[k])
+    for (na, pa), (nb, pb) in zip(pol_a.named_parameters(),
+                                  pol_b.named_parameters()):
+        torch.testing.assert_close(pa, pb, atol=3e-4, rtol=2e-3)
+
+
+@pytest.mark.gpu
+def test_hip_a2c_graph_trains_c4():
+    """config c4 (gandiva+elastic+resize) end-to-end on the HIP A2C path."""
+    import dataclasses
+
+    import numpy as np
+
+    from rlgpuschedule_amd.trainer import CONFIGS, train
+    cfg = dataclasses.replace(CONFIGS["c4_gandiva_a2c"], num_envs=256,
+                              jobs_per_env=32)
+    out = train(cfg, iterations=5, log_fn=None)
+    assert np.isfinite(out["pi_loss"])
+    from rlgpuschedule_amd.agents.a2c_hip import HipA2CLearner  # noqa: F401
