@@ -171,7 +171,7 @@ class HipAdam:
     def step(self):
         if self.grad_reduce is not None:
             self.grad_reduce(self.gflat)
-        norm = torch.linalg.vector_norm(self.gflat)
+        norm = torch.linalg.vector_norm(self.gflat).reshape(1)
         torch.clamp(self.max_norm / (norm + 1e-6), max=1.0, out=self.coef)
         lib = _load_adam()
         st = _stream(self.gflat)
