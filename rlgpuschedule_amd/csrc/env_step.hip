@@ -434,7 +434,7 @@ __device__ bool do_expand(const EnvView& v, Sc& s) {
 
 // Observation + action mask (ENV_SPEC layout). Wave-parallel over slots.
 __device__ void write_obs_mask(const EnvView& v, const Sc& s, int lane) {
-  const int K = (int)v.p->K, N = (int)v.p->N, A = (int)v.p->A;
+  const int K = (int)v.p->K, N = (int)v.p->N;
   const int OD = (int)v.p->obs_dim;
   const int64_t skip = v.p->debug_skip;
   if (skip & 2) return;   // ablation only (bench/abl_env.py)
