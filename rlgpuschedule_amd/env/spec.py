@@ -6,7 +6,7 @@ the spec is our own design built to BASELINE.json's capability list.
 """
 from __future__ import annotations
 
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 
 
 @dataclass(frozen=True)

@@ -8,7 +8,6 @@ here, <10 MB, latency dominates — a single flat buffer beats bucketing).
 from __future__ import annotations
 
 import os
-from contextlib import contextmanager
 
 import torch
 import torch.distributed as dist
