@@ -93,3 +93,29 @@ def test_all_jobs_eventually_complete_under_greedy_schedule(jobs, seed):
             break
     assert done
     assert env.completed == jobs
+
+
+@settings(max_examples=20, deadline=None)
+@given(n=st.integers(4, 64), seed=st.integers(0, 100_000),
+       maxg=st.sampled_from([8, 16, 64]), elastic=st.booleans(),
+       fail=st.sampled_from([0.0, 0.1]))
+def test_trace_generator_wellformed(n, seed, maxg, elastic, fail):
+    """philly_like invariants: sorted arrivals, positive work, demands in
+    [1, max_job_gpus], 1 <= min_gpus <= gpus (== gpus when not elastic)."""
+    tr = philly_like(n, seed=seed, max_job_gpus=maxg, elastic=elastic,
+                     fail_frac=fail)
+    assert len(tr) == n
+    arr = np.asarray(tr.arrival_us)
+    assert (np.diff(arr) >= 0).all()
+    assert (np.asarray(tr.work_us) >= 1).all()
+    g = np.asarray(tr.gpus)
+    mg = np.asarray(tr.min_gpus)
+    assert ((g >= 1) & (g <= maxg)).all()
+    assert ((mg >= 1) & (mg <= g)).all()
+    if not elastic:
+        assert (mg == g).all()
+    # determinism
+    tr2 = philly_like(n, seed=seed, max_job_gpus=maxg, elastic=elastic,
+                      fail_frac=fail)
+    assert np.array_equal(arr, np.asarray(tr2.arrival_us))
+    assert np.array_equal(g, np.asarray(tr2.gpus))
