@@ -70,3 +70,15 @@ def test_cli_eval_policy_cpu(capsys, tmp_path):
                    str(tmp_path / "p.pt"), "--envs", "2", "--episodes", "2",
                    "--jobs-per-env", "10"])
     assert rc == 0
+
+
+def test_quickstart_example_runs(tmp_path, monkeypatch):
+    import subprocess
+    import sys
+    from pathlib import Path
+    root = Path(__file__).resolve().parent.parent
+    r = subprocess.run([sys.executable, str(root / "examples/quickstart.py"),
+                        "--iters", "2"], cwd=tmp_path, capture_output=True,
+                       text=True, timeout=300)
+    assert r.returncode == 0, r.stderr[-800:]
+    assert "ppo: mean JCT" in r.stdout
