@@ -11,8 +11,12 @@ nondeterminism makes late training chaotic), so the reliable recipe is
      its held-out-pool JCT (selection and held-out pools use disjoint
      seeds, so the reported number carries no selection bias).
 
-Single-process by design: a seed sweep is embarrassingly parallel, so
-multi-GPU users run one seed per GPU (see cli.py train-sweep --seeds).
+A seed sweep is embarrassingly parallel: under torch.distributed.run
+(one rank per GPU) train_multi_seed stripes the seed list across ranks
+(rank r trains seeds[r::world]) and all-gathers the per-seed results,
+so `train-sweep --seeds 0 1 2 3 4 5 6 7` on an 8-GPU node trains all
+eight seeds concurrently with no inter-rank communication during
+training.
 """
 from __future__ import annotations
 
@@ -75,7 +79,14 @@ def train_multi_seed(cfg, seeds, iterations: int | None = None,
     number), so `best_heldout_jct_us` is an unbiased estimate of deployed
     quality. cfg.ppo.kl_coef > 0 is strongly recommended (see module doc).
     """
+    import torch.distributed as dist
+
     from ..trainer import train
+
+    rank, world = 0, 1
+    if dist.is_initialized():
+        rank, world = dist.get_rank(), dist.get_world_size()
+    my_seeds = list(seeds)[rank::world] if world > 1 else list(seeds)
 
     device = ("cuda" if torch.cuda.is_available() and cfg.use_hip_env
               else "cpu")
@@ -86,11 +97,11 @@ def train_multi_seed(cfg, seeds, iterations: int | None = None,
            "min_episodes": 16, **(eval_kw or {})}
     runs = []
     best = None
-    for sd in seeds:
+    for sd in my_seeds:
         c = dataclasses.replace(cfg, seed=int(sd))
         sel = BestCheckpointSelector(cfg.spec, select_every, device=device,
                                      **skw)
-        out = train(c, iterations, log_fn=None, callback=sel)
+        out = train(c, iterations, log_fn=None, callback=sel, solo=True)
         policy = out.pop("_policy")
         if sel.state is not None:
             policy.load_state_dict(sel.state)
@@ -109,6 +120,31 @@ def train_multi_seed(cfg, seeds, iterations: int | None = None,
                                       or sel.best_jct < best["jct"]):
             best = {"jct": sel.best_jct, "seed": int(sd),
                     "state": sel.state, "heldout": held["mean_jct_us"]}
+    if world > 1:
+        # gather per-seed records and each rank's best candidate
+        gathered = [None] * world
+        dist.all_gather_object(gathered, {
+            "runs": runs,
+            "best": None if best is None else
+            {k: best[k] for k in ("jct", "seed", "heldout")}})
+        runs = [r for g in gathered for r in g["runs"]]
+        runs.sort(key=lambda r: r["seed"])
+        cands = [g["best"] for g in gathered if g["best"] is not None]
+        if cands:
+            # every rank computes the same winner + source rank, and every
+            # rank joins the broadcast (collective — no conditional paths)
+            win = min(cands, key=lambda c: (c["jct"], c["seed"]))
+            src = next(i for i, g in enumerate(gathered)
+                       if g["best"] is not None
+                       and g["best"]["seed"] == win["seed"])
+            holder = [None]
+            if rank == src:
+                holder = [{k: v.cpu() for k, v in best["state"].items()}]
+            dist.broadcast_object_list(holder, src=src)
+            best = {**win, "state": {
+                k: v.to(device) for k, v in holder[0].items()}}
+        else:
+            best = None
     res = {"runs": runs}
     if best is not None:
         res.update(best_seed=best["seed"],

@@ -99,12 +99,19 @@ def make_learner(cfg: TrainConfig, policy, grad_sync=None):
 
 def train(cfg: TrainConfig, iterations: int | None = None,
           log_fn=print, out_dir: str | None = None, save_every: int = 0,
-          resume: str | None = None, callback=None) -> dict:
+          resume: str | None = None, callback=None,
+          solo: bool = False) -> dict:
     """Synchronous trainer (ppo/a2c). IMPALA lives in agents/impala.py.
     callback(it, policy, stats), if given, runs on rank 0 after every
     update (it must not mutate the policy) — used by agents/selection.py
-    for eval-based checkpoint selection."""
+    for eval-based checkpoint selection.
+    solo=True trains independently even inside an initialized process
+    group (no param broadcast, no grad all-reduce) — used by the striped
+    multi-seed sweep where each rank trains DIFFERENT seeds and matched
+    collective sequences across ranks cannot be guaranteed."""
     rank, world, local = init_from_env()
+    if solo:
+        rank, world = 0, 1
     device = resolve_device(cfg)
     if device.type == "cuda":
         device = torch.device(f"cuda:{local}")
@@ -114,7 +121,8 @@ def train(cfg: TrainConfig, iterations: int | None = None,
         return train_impala(cfg, iterations=iterations, log_fn=log_fn)
     env = make_env(cfg, device, rank)
     policy = make_policy(cfg, device)
-    broadcast_params(policy)
+    if not solo:
+        broadcast_params(policy)
     grad_sync = FlatGradAllreduce(policy) if world > 1 else None
     on_gpu_fast = device.type == "cuda" and cfg.use_hip_env
     if on_gpu_fast:

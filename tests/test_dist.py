@@ -114,3 +114,39 @@ def test_agree_all_gloo():
     for rank, r1, r2 in res:
         assert r1 is False    # one rank voted False -> all False
         assert r2 is True
+
+
+def _worker_sweep(rank, world, port, q):
+    import os
+    os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+                      RANK=str(rank), WORLD_SIZE=str(world),
+                      LOCAL_RANK=str(rank))
+    import dataclasses
+
+    import torch.distributed as d
+    d.init_process_group("gloo", rank=rank, world_size=world)
+    from rlgpuschedule_amd.agents.selection import train_multi_seed
+    from rlgpuschedule_amd.trainer import CONFIGS
+    cfg = dataclasses.replace(
+        CONFIGS["c2_ppo_mlp"], num_envs=4, jobs_per_env=8, rollout_T=8,
+        use_hip_env=False, device="cpu", hidden=16)
+    cfg = dataclasses.replace(
+        cfg, ppo=dataclasses.replace(cfg.ppo, epochs=1, minibatches=1))
+    res = train_multi_seed(
+        cfg, [1, 2, 3], iterations=4, select_every=2,
+        select_kw=dict(num_envs=2, min_episodes=2, max_steps=3000),
+        eval_kw=dict(num_envs=2, min_episodes=2, max_steps=3000))
+    # note: trainer.train calls init_from_env which must not re-init
+    q.put((rank, sorted(r["seed"] for r in res["runs"]), res["best_seed"],
+           float(sum(v.sum() for v in res["best_state"].values()))))
+    d.destroy_process_group()
+
+
+def test_multi_seed_sweep_striped_across_ranks_gloo():
+    """3 seeds striped over 2 gloo ranks: every rank sees all 3 results,
+    agrees on the winner, and holds the identical winning state."""
+    res = _spawn(_worker_sweep)
+    (r0, seeds0, best0, sum0), (r1, seeds1, best1, sum1) = res
+    assert seeds0 == seeds1 == [1, 2, 3]
+    assert best0 == best1
+    assert abs(sum0 - sum1) < 1e-6
