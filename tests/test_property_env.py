@@ -15,26 +15,31 @@ from rlgpuschedule_amd.traces.synthetic import philly_like
 
 
 def _mk_env(nodes, gpn, gandiva, elastic, jobs, seed, obs_version=1,
-            resize=False):
+            resize=False, K=16, P=256, S=8):
     spec = EnvSpec(num_nodes=nodes, gpus_per_node=gpn, horizon=80_000,
                    gandiva=gandiva, elastic=elastic, obs_version=obs_version,
-                   elastic_resize=resize and elastic)
+                   elastic_resize=resize and elastic, queue_visible=K,
+                   queue_cap=P, max_span=S)
     tr = philly_like(jobs, seed=seed, max_job_gpus=spec.max_job_gpus,
                      mean_interarrival_s=40.0, median_duration_s=300.0,
                      elastic=elastic)
     return spec, ClusterEnvRef(spec, tr)
 
 
-@settings(max_examples=30, deadline=None)
+@settings(max_examples=60, deadline=None)
 @given(nodes=st.integers(1, 6), gpn=st.sampled_from([2, 4, 8]),
        gandiva=st.booleans(), elastic=st.booleans(),
        jobs=st.integers(4, 48), seed=st.integers(0, 10_000),
        aseed=st.integers(0, 10_000), obs_version=st.sampled_from([1, 2]),
-       resize=st.booleans())
+       resize=st.booleans(), K=st.sampled_from([2, 8, 16, 32]),
+       P=st.sampled_from([8, 64, 256]), S=st.sampled_from([1, 2, 8]))
 def test_invariants_hold_under_random_masked_actions(
-        nodes, gpn, gandiva, elastic, jobs, seed, aseed, obs_version, resize):
+        nodes, gpn, gandiva, elastic, jobs, seed, aseed, obs_version, resize,
+        K, P, S):
+    if K > P:
+        K = P
     spec, env = _mk_env(nodes, gpn, gandiva, elastic, jobs, seed,
-                        obs_version, resize)
+                        obs_version, resize, K=K, P=P, S=S)
     env.reset()
     rng = np.random.default_rng(aseed)
     for t in range(200):
