@@ -108,7 +108,8 @@ def train_multi_seed(cfg, seeds, iterations: int | None = None,
         held = evaluate_policy(cfg.spec, policy, greedy=True, device=device,
                                **ekw)
         rec = {"seed": int(sd), "selected_at": sel.at,
-               "selection_jct_us": sel.best_jct,
+               "selection_jct_us": (sel.best_jct if sel.state is not None
+                                    else None),  # None: selection disabled
                "heldout_jct_us": held["mean_jct_us"],
                "heldout_episodes": held["episodes"]}
         runs.append(rec)

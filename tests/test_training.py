@@ -182,3 +182,23 @@ def test_cli_train_sweep_cpu(tmp_path):
                "--select-every", "1",
                "--save-best", str(tmp_path / "best.pt")])
     assert rc == 0
+
+
+def test_multi_seed_selection_disabled_is_json_safe():
+    """select_every=0: no snapshots, result must serialize as strict JSON
+    (no Infinity) and carry the final policies' held-out evals."""
+    import dataclasses
+    import json
+
+    from rlgpuschedule_amd.agents.selection import train_multi_seed
+    from rlgpuschedule_amd.trainer import CONFIGS
+    cfg = dataclasses.replace(
+        CONFIGS["c2_ppo_mlp"], num_envs=2, jobs_per_env=8, rollout_T=8,
+        use_hip_env=False, device="cpu", hidden=16)
+    cfg = dataclasses.replace(
+        cfg, ppo=dataclasses.replace(cfg.ppo, epochs=1, minibatches=1))
+    res = train_multi_seed(cfg, [1], iterations=2, select_every=0,
+                           eval_kw=dict(num_envs=2, min_episodes=2,
+                                        max_steps=3000))
+    assert "best_seed" not in res
+    json.dumps(res, allow_nan=False)   # raises on inf/nan
