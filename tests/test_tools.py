@@ -28,3 +28,20 @@ def test_rocpd_to_kernel_stats_roundtrip(tmp_path):
     lines = out.read_text().strip().splitlines()
     assert lines[0].startswith('"Name"')
     assert len(lines) == 3
+
+
+def test_survey_reference_scanner(tmp_path):
+    from tools.survey_reference import scan
+    (tmp_path / "src").mkdir()
+    (tmp_path / "src/env.py").write_text(
+        "class Env:\n    def step(self, a):\n        pass\n"
+        "    def reset(self):\n        pass\n")
+    (tmp_path / "src/train.cu").write_text("__global__ void k() {}\n")
+    out = scan(tmp_path)
+    assert "2 files" in out
+    assert "src/env.py:2" in out
+    assert "__global__" in out
+    # empty dir
+    empty = tmp_path / "empty"
+    empty.mkdir()
+    assert "EMPTY" in scan(empty)
