@@ -79,6 +79,6 @@ def test_quickstart_example_runs(tmp_path, monkeypatch):
     root = Path(__file__).resolve().parent.parent
     r = subprocess.run([sys.executable, str(root / "examples/quickstart.py"),
                         "--iters", "2"], cwd=tmp_path, capture_output=True,
-                       text=True, timeout=300)
+                       text=True, timeout=900)
     assert r.returncode == 0, r.stderr[-800:]
     assert "ppo: mean JCT" in r.stdout
