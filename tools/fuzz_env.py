@@ -60,5 +60,8 @@ while time.time() - t0 < budget:
         if done:
             break
     n_cfg += 1
+    if n_cfg % 20000 == 0:
+        print(f"... {n_cfg} configs clean, {time.time()-t0:.0f}s",
+              flush=True)
 print(f"FUZZ CLEAN: {n_cfg} random configs x <=300 steps, "
       f"{time.time()-t0:.0f}s")
