@@ -152,8 +152,9 @@ def main():
             "vs_baseline": None,  # no published reference numbers
             "dtype": (f"mixed (actor {args.actor_dtype}, "
                       f"update {args.update_dtype})"
-                      if "bf16" in (args.actor_dtype, args.update_dtype)
-                      else "fp32"),
+                      if has_gpu and "bf16" in (args.actor_dtype,
+                                                args.update_dtype)
+                      else "fp32"),   # CPU fallback is always torch fp32
             "data": "synthetic",
             "mean_jct_us": round(st.get("mean_jct_us", 0.0), 1),
             "episodes": st.get("episodes", 0),
@@ -166,7 +167,7 @@ def main():
                 "envs_per_gpu": E,
                 "ppo": f"1 epoch x {args.minibatches} minibatches, "
                        "clip 0.2, GAE(0.99, 0.95), reward-norm",
-                "update_path": args.learner,
+                "update_path": args.learner if has_gpu else "torch-cpu",
                 "parallelism": f"dp{n_gpus}",
                 "env_shards": args.shards if has_gpu else 1,
             },
