@@ -47,6 +47,14 @@ class VtraceLearner:
         self.grad_sync = grad_sync
         self.opt = torch.optim.Adam(policy.parameters(), lr=cfg.lr, eps=1e-5)
 
+    def load_optim_state(self, sd):
+        if isinstance(sd, dict) and sd.get("hip_adam"):
+            import warnings
+            warnings.warn("checkpoint carries HipAdam state but this is the "
+                          "torch learner; optimizer state not restored")
+            return
+        self.opt.load_state_dict(sd)
+
     def update(self, buf: RolloutBuffer) -> dict:
         cfg = self.cfg
         if cfg.norm_reward:
@@ -89,8 +97,13 @@ class VtraceLearner:
                 "entropy": ent.item(), "mean_reward": buf.reward.mean().item()}
 
 
-def train_impala(cfg, iterations=None, log_fn=print) -> dict:
-    """Async actor/learner loop (cfg is a trainer.TrainConfig)."""
+def train_impala(cfg, iterations=None, log_fn=print, out_dir=None,
+                 save_every=0, resume=None) -> dict:
+    """Async actor/learner loop (cfg is a trainer.TrainConfig).
+
+    out_dir/save_every/resume mirror trainer.train() so the c5_impala
+    config gets checkpointing, metrics logging and resume too (ADVICE r1:
+    these silently no-op'd before)."""
     from ..parallel.dist import (FlatGradAllreduce, broadcast_params,
                                  init_from_env)
     from ..trainer import make_env, make_policy, resolve_device
@@ -109,6 +122,32 @@ def train_impala(cfg, iterations=None, log_fn=print) -> dict:
         learner = HipVtraceLearner(policy, icfg, grad_sync)
     else:
         learner = VtraceLearner(policy, icfg, grad_sync)
+    start_it = 0
+    if resume:
+        from ..ckpt.io import apply_optim_state, load_policy
+        meta = load_policy(resume, policy, map_location=str(device))
+        apply_optim_state(learner, meta.get("optim_state"))
+        start_it = int(meta.get("meta", {}).get("iteration", 0))
+        if rank == 0 and log_fn:
+            log_fn(f"[{cfg.name}] resumed from {resume} at iter {start_it}")
+    logger = None
+    if out_dir is not None and rank == 0:
+        from ..metrics.logger import MetricsLogger
+        logger = MetricsLogger(out_dir, echo=False,
+                               env_steps_per_iter=cfg.num_envs * cfg.rollout_T)
+
+    def after_update(u, stats, last):
+        """Rank-0 bookkeeping after completed update index u (0-based,
+        resume-global)."""
+        if logger is not None:
+            logger.log(stats, iteration=u)
+        if (save_every and rank == 0 and out_dir is not None
+                and ((u + 1) % save_every == 0 or last)):
+            from ..ckpt.io import save_policy
+            save_policy(f"{out_dir}/policy_{u + 1:06d}.pt", policy,
+                        spec=cfg.spec, optimizer=getattr(learner, "opt", None),
+                        meta={"iteration": u + 1, "config": cfg.name})
+
     iters = iterations if iterations is not None else cfg.iterations
     stats = {}
 
@@ -146,6 +185,7 @@ def train_impala(cfg, iterations=None, log_fn=print) -> dict:
                 if rank == 0 and log_fn and (it % 10 == 1 or it == iters):
                     log_fn(f"[impala] update {it}: " +
                            " ".join(f"{k}={v:.4f}" for k, v in stats.items()))
+                after_update(start_it + it - 1, stats, it == iters)
         out = dict(stats)
         out["episode_stats"] = env.episode_stats()
         return out
@@ -154,14 +194,15 @@ def train_impala(cfg, iterations=None, log_fn=print) -> dict:
     behavior = copy.deepcopy(policy)
     collector = Collector(env, behavior, cfg.rollout_T, device,
                           seed=cfg.seed * 101 + rank)
-    for it in range(iters):
+    for it in range(start_it, start_it + iters):
         if it % icfg.behavior_sync == 0:
             behavior.load_state_dict(policy.state_dict())
         buf = collector.collect()
         stats = learner.update(buf)
-        if rank == 0 and log_fn and (it % 10 == 0 or it == iters - 1):
+        if rank == 0 and log_fn and (it % 10 == 0 or it == start_it + iters - 1):
             log_fn(f"[impala-cpu] iter {it}: " +
                    " ".join(f"{k}={v:.4f}" for k, v in stats.items()))
+        after_update(it, stats, it == start_it + iters - 1)
     out = dict(stats)
     if hasattr(env, "venv"):
         st = env.venv.stats

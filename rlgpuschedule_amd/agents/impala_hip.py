@@ -91,6 +91,13 @@ class HipVtraceLearner:
                            max_norm=self.cfg.max_grad_norm,
                            grad_reduce=allreduce_flat
                            if self.grad_sync is not None else None)
+        from ..ckpt.io import resolve_pending_opt_state
+        resolve_pending_opt_state(self)
+
+    def load_optim_state(self, sd):
+        """Stash checkpointed optimizer state; applied when _alloc creates
+        the real optimizer (HipAdam is lazy — see ckpt.io)."""
+        self._pending_opt_state = sd
 
     def _update_body(self, buf: RolloutBuffer):
         cfg = self.cfg

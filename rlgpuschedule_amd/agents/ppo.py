@@ -53,6 +53,14 @@ class PPOLearner:
         self.opt = torch.optim.Adam(policy.parameters(), lr=cfg.lr, eps=1e-5)
         self.grad_sync = grad_sync  # parallel/dist.py hook (RCCL all-reduce)
 
+    def load_optim_state(self, sd):
+        if isinstance(sd, dict) and sd.get("hip_adam"):
+            import warnings
+            warnings.warn("checkpoint carries HipAdam state but this is the "
+                          "torch learner; optimizer state not restored")
+            return
+        self.opt.load_state_dict(sd)
+
     def update(self, buf: RolloutBuffer) -> dict:
         cfg = self.cfg
         if cfg.norm_reward:
@@ -181,6 +189,8 @@ class A2CLearner:
         self.opt = torch.optim.RMSprop(policy.parameters(), lr=cfg.lr,
                                        alpha=0.99, eps=1e-5)
         self.grad_sync = grad_sync
+
+    load_optim_state = PPOLearner.load_optim_state
 
     def update(self, buf: RolloutBuffer) -> dict:
         cfg = self.cfg

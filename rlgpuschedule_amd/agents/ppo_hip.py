@@ -131,6 +131,13 @@ class HipPPOLearner:
                 grad_reduce=allreduce_flat if self.grad_sync is not None
                 else None)
             self.opt = self.hip_opt
+        from ..ckpt.io import resolve_pending_opt_state
+        resolve_pending_opt_state(self)
+
+    def load_optim_state(self, sd):
+        """Stash checkpointed optimizer state; applied when _alloc creates
+        the real optimizer (HipAdam is lazy — see ckpt.io)."""
+        self._pending_opt_state = sd
 
     def _refill_perm(self):
         for ep in range(self.cfg.epochs):

@@ -191,12 +191,27 @@ class HipAdam:
 
     # torch-optimizer duck typing for ckpt/io.py
     def state_dict(self):
-        return {"hip_adam": True, "m": self.m, "v": self.v, "t": self.t}
+        return {"hip_adam": True, "m": self.m, "v": self.v, "t": self.t,
+                "bc": self.bc}
 
     def load_state_dict(self, sd):
+        if not (isinstance(sd, dict) and sd.get("hip_adam")):
+            raise ValueError(
+                "HipAdam.load_state_dict got non-HipAdam optimizer state "
+                "(e.g. torch Adam); refusing to restore silently — use "
+                "ckpt.io.apply_optim_state for cross-path dispatch")
         self.m.copy_(sd["m"].to(self.m.device))
         self.v.copy_(sd["v"].to(self.v.device))
         self.t.copy_(sd["t"].to(self.t.device))
+        if "bc" in sd:
+            self.bc.copy_(sd["bc"].to(self.bc.device))
+        else:
+            # legacy checkpoint without the running beta^t products:
+            # recompute so bias correction stays right after resume
+            t = float(sd["t"].reshape(-1)[0].item())
+            self.bc.copy_(torch.tensor(
+                [self.betas[0] ** t, self.betas[1] ** t],
+                device=self.bc.device))
 
 
 def _load_bf16():

@@ -117,8 +117,13 @@ def train(cfg: TrainConfig, iterations: int | None = None,
         device = torch.device(f"cuda:{local}")
     torch.manual_seed(cfg.seed * 31 + rank)
     if cfg.algo == "impala":
+        if callback is not None or solo:
+            raise ValueError("algo='impala' does not support callback/solo "
+                             "(selection pipelines run on the sync trainer)")
         from .agents.impala import train_impala
-        return train_impala(cfg, iterations=iterations, log_fn=log_fn)
+        return train_impala(cfg, iterations=iterations, log_fn=log_fn,
+                            out_dir=out_dir, save_every=save_every,
+                            resume=resume)
     env = make_env(cfg, device, rank)
     policy = make_policy(cfg, device)
     if not solo:
@@ -151,10 +156,12 @@ def train(cfg: TrainConfig, iterations: int | None = None,
         learner = make_learner(cfg, policy, grad_sync)
     start_it = 0
     if resume:
-        from .ckpt.io import load_policy
-        meta = load_policy(resume, policy,
-                           getattr(learner, "opt", None),
-                           map_location=str(device))
+        from .ckpt.io import apply_optim_state, load_policy
+        # Do NOT pass the optimizer here: the HIP learners create HipAdam
+        # lazily in _alloc, so the state is routed via load_optim_state
+        # and applied once the real optimizer exists.
+        meta = load_policy(resume, policy, map_location=str(device))
+        apply_optim_state(learner, meta.get("optim_state"))
         start_it = int(meta.get("meta", {}).get("iteration", 0))
         if rank == 0 and log_fn:
             log_fn(f"[{cfg.name}] resumed from {resume} at iter {start_it}")
