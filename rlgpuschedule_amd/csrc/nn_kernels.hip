@@ -1033,3 +1033,14 @@ extern "C" int rlsched_gru_bf16(const int64_t* ip, const uint64_t* bufs,
                      (hipStream_t)stream, p);
   return (int)hipGetLastError();
 }
+
+// Build provenance: ops/build.py embeds a sha256-prefix of this lib's
+// sources at compile time and verifies it at load (stale-binary guard,
+// VERDICT r1 weak #6). The "RLSHASH:" marker makes the hash findable by
+// byte-scan without dlopen.
+#ifndef RLS_SOURCE_HASH
+#define RLS_SOURCE_HASH "unset"
+#endif
+extern "C" const char* rlsched_source_hash() {
+  return "RLSHASH:" RLS_SOURCE_HASH;
+}

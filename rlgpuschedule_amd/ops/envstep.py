@@ -10,7 +10,7 @@ import ctypes
 
 import numpy as np
 
-from .build import build_lib
+from .build import build_lib, verify_loaded
 
 BUF_ORDER = [
     "trace_arr", "trace_gpus", "trace_work", "trace_ming",
@@ -32,6 +32,7 @@ def load() -> ctypes.CDLL:
     if _lib is None:
         so = build_lib("librlsched_env.so")
         lib = ctypes.CDLL(str(so))
+        verify_loaded(lib, "librlsched_env.so")
         for fn in ("rlsched_env_reset", "rlsched_env_step"):
             f = getattr(lib, fn)
             f.restype = ctypes.c_int

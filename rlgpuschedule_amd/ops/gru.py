@@ -6,7 +6,7 @@ import ctypes
 import numpy as np
 import torch
 
-from .build import build_lib
+from .build import build_lib, verify_loaded
 
 _lib = None
 
@@ -16,6 +16,7 @@ def load() -> ctypes.CDLL:
     if _lib is None:
         so = build_lib("librlsched_gru.so")
         lib = ctypes.CDLL(str(so))
+        verify_loaded(lib, "librlsched_gru.so")
         for fn in ("rlsched_gru_seq_fwd", "rlsched_gru_seq_bwd",
                    "rlsched_head_fwd"):
             f = getattr(lib, fn)

@@ -7,7 +7,7 @@ import ctypes
 import numpy as np
 import torch
 
-from .build import build_lib
+from .build import build_lib, verify_loaded
 
 _lib = None
 
@@ -17,6 +17,7 @@ def load() -> ctypes.CDLL:
     if _lib is None:
         so = build_lib("librlsched_nn.so")
         lib = ctypes.CDLL(str(so))
+        verify_loaded(lib, "librlsched_nn.so")
         lib.rlsched_gae.restype = ctypes.c_int
         lib.rlsched_gae.argtypes = [ctypes.c_void_p] * 4
         lib.rlsched_actor_mlp.restype = ctypes.c_int

@@ -8,7 +8,7 @@ import ctypes
 import numpy as np
 import torch
 
-from .build import build_lib
+from .build import build_lib, verify_loaded
 
 _lib = None
 
@@ -18,6 +18,7 @@ def load() -> ctypes.CDLL:
     if _lib is None:
         so = build_lib("librlsched_train.so")
         lib = ctypes.CDLL(str(so))
+        verify_loaded(lib, "librlsched_train.so")
         for fn, na in (("rlsched_fwd_train", 3), ("rlsched_ppo_head", 4),
                        ("rlsched_bwd_dx", 3), ("rlsched_bwd_dw", 3),
                        ("rlsched_colsum", 3)):
