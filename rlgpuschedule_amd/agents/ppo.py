@@ -44,6 +44,11 @@ class PPOConfig:
     # collapse documented in docs/PERF.md (policy drifts far off-behavior
     # between graph-captured updates).
     kl_coef: float = 0.0
+    # Entropy floor (round 2): samples whose per-row policy entropy is
+    # below ent_floor (nats) get an EXTRA entropy bonus of ent_floor_coef
+    # on top of ent_coef — exploration is forced only where it collapsed.
+    ent_floor: float = 0.0
+    ent_floor_coef: float = 0.0
 
 
 class PPOLearner:
@@ -93,6 +98,12 @@ class PPOLearner:
         v_loss = 0.5 * torch.max((value - ret) ** 2, (v_clip - ret) ** 2).mean()
         ent = entropy.mean()
         loss = pi_loss + cfg.vf_coef * v_loss - cfg.ent_coef * ent
+        if getattr(cfg, "ent_floor_coef", 0.0):
+            # per-row hinge: extra bonus only where entropy collapsed
+            # (matches the fused ppo_head_kernel; the hinge gate is
+            # treated as a constant wrt the gradient, like the kernel)
+            gate = (entropy.detach() < cfg.ent_floor).float()
+            loss = loss - cfg.ent_floor_coef * (gate * entropy).mean()
         stats["pi_loss"] += pi_loss.item()
         stats["v_loss"] += v_loss.item()
         stats["entropy"] += ent.item()

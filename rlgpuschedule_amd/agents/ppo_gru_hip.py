@@ -197,7 +197,9 @@ class HipGRUPPOLearner:
                          s["ov"].view(B), s["advm"].reshape(B),
                          s["retm"].view(B), cfg.clip, cfg.vf_coef,
                          cfg.ent_coef, s["dhead"], s["stats"],
-                         kl_coef=getattr(cfg, "kl_coef", 0.0))
+                         kl_coef=getattr(cfg, "kl_coef", 0.0),
+                         ent_floor=getattr(cfg, "ent_floor", 0.0),
+                         ent_floor_coef=getattr(cfg, "ent_floor_coef", 0.0))
                 if self.dtype == "bf16":
                     bwd_dx_bf16(s["dhead"], mi.whT, None,
                                 s["dh_ext"].view(B, H))

@@ -185,7 +185,9 @@ class HipPPOLearner:
                 ppo_head(s["head"], s["m"], s["a"], s["olp"], s["ov"],
                          s["advm"], s["retm"], cfg.clip, cfg.vf_coef,
                          cfg.ent_coef, s["dhead"], s["stats"],
-                         kl_coef=getattr(cfg, "kl_coef", 0.0))
+                         kl_coef=getattr(cfg, "kl_coef", 0.0),
+                         ent_floor=getattr(cfg, "ent_floor", 0.0),
+                         ent_floor_coef=getattr(cfg, "ent_floor_coef", 0.0))
                 if self.dtype == "bf16":
                     bwd_dx_bf16(s["dhead"], mi.whT, s["h2"], s["dh2"])
                     bwd_dx_bf16(s["dh2"], mi.w2T, s["h1"], s["dh1"])

@@ -116,6 +116,12 @@ __global__ __launch_bounds__(256) void fwd_train_kernel(FwdTrainArgs p) {
 struct PpoHeadArgs {
   int64_t B, A;
   float clip, vf_coef, ent_coef, inv_n, kl_coef;
+  // Entropy floor (round 2): rows whose policy entropy has
+  // collapsed below ent_floor get an EXTRA entropy bonus of
+  // ent_floor_coef on top of ent_coef. A per-row hinge — exploration is
+  // only forced where it has died, instead of globally inflating
+  // ent_coef (which round-1 sweeps showed destabilizes training).
+  float ent_floor, ent_floor_coef;
   const float* head;       // [B,HEADC]
   const uint8_t* mask;     // [B,A]
   const int64_t* actions;  // [B]
@@ -166,13 +172,15 @@ __global__ __launch_bounds__(256) void ppo_head_kernel(PpoHeadArgs a) {
     s_cf += (fabsf(ratio - 1.f) > a.clip) ? 1.f : 0.f;
     const float g_logp = (-advi * ratio * active
                           + a.kl_coef * (ratio - 1.f)) * a.inv_n;
+    const float ecoef = a.ent_coef
+        + ((H < a.ent_floor) ? a.ent_floor_coef : 0.f);
     for (int j = 0; j < A; ++j) {
       float g = 0.f;
       if (mk[j]) {
         float lpj = (lg[j] - mx) - lsum;
         float pj = __expf(lpj);
         g = g_logp * (((j == act) ? 1.f : 0.f) - pj)
-          + a.ent_coef * pj * (lpj + H) * a.inv_n;
+          + ecoef * pj * (lpj + H) * a.inv_n;
       }
       dlg[j] = g;
     }
@@ -397,6 +405,7 @@ extern "C" int rlsched_ppo_head(const int64_t* ip, const double* fp,
   a.clip = (float)fp[0]; a.vf_coef = (float)fp[1];
   a.ent_coef = (float)fp[2]; a.inv_n = (float)fp[3];
   a.kl_coef = (float)fp[4];
+  a.ent_floor = (float)fp[5]; a.ent_floor_coef = (float)fp[6];
   a.head = (const float*)bufs[0];
   a.mask = (const uint8_t*)bufs[1];
   a.actions = (const int64_t*)bufs[2];

@@ -58,14 +58,18 @@ def fwd_train(obs, wt1, b1, wt2, b2, wth, bh, h1, h2, head):
 
 
 def ppo_head(head, mask, actions, old_logp, old_value, adv, ret,
-             clip, vf_coef, ent_coef, dhead, stats, kl_coef=0.0):
+             clip, vf_coef, ent_coef, dhead, stats, kl_coef=0.0,
+             ent_floor=0.0, ent_floor_coef=0.0):
     """Masked PPO loss on head rows; writes dhead + accumulates stats[4].
-    kl_coef>0 adds the k3 KL-to-behavior penalty (folded into pi stat)."""
+    kl_coef>0 adds the k3 KL-to-behavior penalty (folded into pi stat).
+    ent_floor/ent_floor_coef: rows with entropy < ent_floor get an extra
+    entropy bonus (per-row hinge against entropy collapse)."""
     B, A = mask.shape
     _call("rlsched_ppo_head", [B, A],
           _ptrs(head, mask, actions, old_logp, old_value, adv, ret,
                 dhead, stats),
-          _stream(head), fp=[clip, vf_coef, ent_coef, 1.0 / B, kl_coef])
+          _stream(head), fp=[clip, vf_coef, ent_coef, 1.0 / B, kl_coef,
+                             ent_floor, ent_floor_coef])
 
 
 def bwd_dx(dY, wtp, h, dX):

@@ -284,6 +284,35 @@ def test_hip_ppo_kl_penalty_matches_reference():
 
 
 @pytest.mark.gpu
+def test_hip_ppo_entropy_floor_matches_reference():
+    """ent_floor/ent_floor_coef (per-row hinge entropy bonus) parity vs
+    torch autograd reference. Floor is set mid-range so the gate is
+    exercised in BOTH states across rows (row entropies vary with the
+    action-mask population)."""
+    from helpers import make_buf as _make_buf
+    from rlgpuschedule_amd.agents.ppo import PPOConfig, PPOLearner
+    from rlgpuschedule_amd.agents.ppo_hip import HipPPOLearner
+    from rlgpuschedule_amd.models.policy import MLPPolicy
+    torch.manual_seed(13)
+    spec = cluster64()
+    T, E = 16, 64
+    pol_a = MLPPolicy(spec.obs_dim, spec.num_actions, 256).cuda()
+    pol_b = copy.deepcopy(pol_a)
+    cfg = PPOConfig(epochs=1, minibatches=1, norm_reward=False,
+                    ent_floor=2.0, ent_floor_coef=0.05)
+    ref = PPOLearner(pol_a, cfg)
+    hip = HipPPOLearner(pol_b, cfg, use_graph=False)
+    buf = _make_buf(T, E, spec, "cuda", seed=21)
+    s_ref = ref.update(buf)
+    s_hip = hip.update(buf)
+    for k in ("pi_loss", "v_loss", "entropy", "clipfrac"):
+        assert abs(s_ref[k] - s_hip[k]) < 5e-3, (k, s_ref[k], s_hip[k])
+    for (na, pa), (nb, pb) in zip(pol_a.named_parameters(),
+                                  pol_b.named_parameters()):
+        torch.testing.assert_close(pa, pb, atol=3e-4, rtol=1e-3)
+
+
+@pytest.mark.gpu
 def test_hip_a2c_learner_matches_reference():
     """HipA2CLearner (fused head + RMSprop-mode step) vs torch A2CLearner."""
     import copy
