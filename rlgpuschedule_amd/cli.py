@@ -27,6 +27,9 @@ def main(argv=None):
     tr.add_argument("--resume", default=None, help="checkpoint path")
     tr.add_argument("--actor-dtype", choices=["fp32", "bf16"], default=None)
     tr.add_argument("--update-dtype", choices=["fp32", "bf16"], default=None)
+    tr.add_argument("--trace-file", default=None,
+                    help="replay a real Philly-style CSV trace "
+                         "(schema: traces/philly.py)")
     tr.add_argument("--obs-version", type=int, choices=[1, 2], default=None,
                     help="2 = richer observations (docs/ENV_SPEC.md v4)")
     ev = sub.add_parser("eval", help="evaluate a checkpoint or baseline")
@@ -38,6 +41,8 @@ def main(argv=None):
     ev.add_argument("--episodes", type=int, default=32)
     ev.add_argument("--jobs-per-env", type=int, default=64)
     ev.add_argument("--seed", type=int, default=1234)
+    ev.add_argument("--trace-file", default=None,
+                    help="evaluate on a real Philly-style CSV trace")
     ev.add_argument("--sample", action="store_true",
                     help="sample instead of greedy argmax")
     sw = sub.add_parser("train-sweep", help="multi-seed training with "
@@ -66,6 +71,10 @@ def main(argv=None):
         import json
         from .agents.evaluate import evaluate_baseline, evaluate_policy
         cfg = CONFIGS[args.config]
+        if args.trace_file:
+            cfg = dataclasses.replace(
+                cfg, trace_kind="file",
+                trace_kw={"path": args.trace_file})
         if args.baseline:
             st = evaluate_baseline(cfg.spec, args.baseline, args.envs,
                                    args.episodes,
@@ -125,6 +134,9 @@ def main(argv=None):
     if args.obs_version is not None:
         over["spec"] = dataclasses.replace(cfg.spec,
                                            obs_version=args.obs_version)
+    if args.trace_file is not None:
+        over["trace_kind"] = "file"
+        over["trace_kw"] = {"path": args.trace_file}
     if over:
         cfg = dataclasses.replace(cfg, **over)
     out = train(cfg, iterations=args.iterations, out_dir=args.out,

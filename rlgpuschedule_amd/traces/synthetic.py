@@ -119,7 +119,15 @@ def make_pool(kind: str, num_envs: int, jobs_per_env: int, seed: int,
     then concatenated (each window is itself arrival-sorted; windows are
     independent, arrivals are re-based per window by the env).
     Returns (pool, job_lo[num_envs], job_hi[num_envs]).
+
+    kind="file": replay a REAL trace file (CSV schema in traces/philly.py)
+    — kw must carry path=...; optional oversize="clip"|"drop".
     """
+    if kind == "file":
+        from .philly import load_trace_csv, windows_from_trace
+        path = kw.pop("path")
+        tr = load_trace_csv(path, max_job_gpus, **kw)
+        return windows_from_trace(tr, num_envs, jobs_per_env, seed)
     gen = philly_like if kind == "philly" else poisson
     arrs, gps, wks, mgs, lo, hi = [], [], [], [], [], []
     off = 0
