@@ -35,6 +35,15 @@ class ImpalaConfig:
     max_grad_norm: float = 0.5
     behavior_sync: int = 1   # refresh behavior policy every N updates
     norm_reward: bool = True
+    # Ablation switch: False disables the V-trace
+    # off-policy correction — targets become plain GAE computed as if the
+    # rollout were on-policy (no rho/c importance clipping). Used by
+    # bench/impala_staleness.py to show the correction earns its keep as
+    # behavior_sync (staleness) grows.
+    use_vtrace: bool = True
+    # Ablation aid: run the torch VtraceLearner even where the HIP one
+    # would be picked, so both ablation arms share one implementation.
+    force_torch_learner: bool = False
 
 
 class VtraceLearner:
@@ -71,7 +80,14 @@ class VtraceLearner:
         lp_cur = lp_cur.reshape(T, E)
         with torch.no_grad():
             v_det = values.detach()
-            if buf.reward.is_cuda:
+            if not cfg.use_vtrace:
+                # ablation: pretend the rollout is on-policy (plain GAE)
+                from ..ops.gae_ref import gae_advantages
+                pg_adv, ret = gae_advantages(
+                    buf.reward, v_det, buf.done, buf.last_value,
+                    cfg.gamma, cfg.lam)
+                vs = ret
+            elif buf.reward.is_cuda:
                 from ..ops.nn import vtrace_scan
                 vs, pg_adv = vtrace_scan(
                     buf.reward, v_det, buf.done, buf.last_value,
@@ -98,7 +114,8 @@ class VtraceLearner:
 
 
 def train_impala(cfg, iterations=None, log_fn=print, out_dir=None,
-                 save_every=0, resume=None) -> dict:
+                 save_every=0, resume=None, icfg: ImpalaConfig | None = None
+                 ) -> dict:
     """Async actor/learner loop (cfg is a trainer.TrainConfig).
 
     out_dir/save_every/resume mirror trainer.train() so the c5_impala
@@ -112,15 +129,18 @@ def train_impala(cfg, iterations=None, log_fn=print, out_dir=None,
     if device.type == "cuda":
         device = torch.device(f"cuda:{local}")
     torch.manual_seed(cfg.seed * 31 + rank)
-    icfg = ImpalaConfig()
+    icfg = icfg if icfg is not None else ImpalaConfig()
     env = make_env(cfg, device, rank)
     policy = make_policy(cfg, device)
     broadcast_params(policy)
     grad_sync = FlatGradAllreduce(policy) if world > 1 else None
-    if device.type == "cuda" and not policy.is_recurrent and cfg.use_hip_env:
+    if (device.type == "cuda" and not policy.is_recurrent
+            and cfg.use_hip_env and icfg.use_vtrace
+            and not icfg.force_torch_learner):
         from .impala_hip import HipVtraceLearner
         learner = HipVtraceLearner(policy, icfg, grad_sync)
     else:
+        # torch learner (also carries the use_vtrace=False ablation path)
         learner = VtraceLearner(policy, icfg, grad_sync)
     start_it = 0
     if resume:

@@ -107,3 +107,61 @@ def test_impala_gpu_hip_learner_pipeline():
         CONFIGS["c5_impala"], num_envs=512, jobs_per_env=32, rollout_T=16)
     out = train(cfg, iterations=6, log_fn=None)
     assert np.isfinite(out["pi_loss"])
+
+
+def test_vtrace_correction_engages_under_staleness():
+    """with behavior_sync=4 the rollout is genuinely
+    off-policy (stored behavior logp != current policy logp) and the
+    V-trace arm's update differs from the plain-GAE (use_vtrace=False)
+    arm; with behavior_sync=1 (on-policy, lam=1) the two coincide."""
+    import copy
+    import dataclasses
+
+    import torch
+
+    from rlgpuschedule_amd.agents.impala import ImpalaConfig, VtraceLearner
+    from rlgpuschedule_amd.agents.rollout import Collector
+    from rlgpuschedule_amd.env.spec import cluster16
+    from rlgpuschedule_amd.models.policy import evaluate_actions
+    from rlgpuschedule_amd.trainer import CONFIGS, make_env, make_policy
+
+    cfg = dataclasses.replace(
+        CONFIGS["c5_impala"], spec=cluster16(), num_envs=8, jobs_per_env=16,
+        rollout_T=16, use_hip_env=False, device="cpu", hidden=32, seed=7)
+    dev = torch.device("cpu")
+
+    def run(sync, vtrace, iters=8):
+        torch.manual_seed(123)
+        icfg = ImpalaConfig(behavior_sync=sync, use_vtrace=vtrace)
+        env = make_env(cfg, dev, 0)
+        pol = make_policy(cfg, dev)
+        lr = VtraceLearner(pol, icfg)
+        behavior = copy.deepcopy(pol)
+        col = Collector(env, behavior, cfg.rollout_T, dev, seed=1)
+        stale = 0.0
+        for it in range(iters):
+            if it % icfg.behavior_sync == 0:
+                behavior.load_state_dict(pol.state_dict())
+            buf = col.collect()
+            with torch.no_grad():
+                T, E = buf.reward.shape
+                logits, _, _ = pol(buf.obs.reshape(T * E, -1),
+                                   buf.mask.reshape(T * E, -1), None)
+                lp_cur, _ = evaluate_actions(logits,
+                                             buf.actions.reshape(T * E))
+                stale = max(stale, float(
+                    (lp_cur - buf.logp.reshape(T * E)).abs().max()))
+            lr.update(buf)
+        psum = float(sum(p.detach().sum() for p in pol.parameters()))
+        return stale, psum
+
+    # staleness is real at sync=4
+    stale4_vt, p4_vt = run(4, True)
+    assert stale4_vt > 1e-4, "behavior_sync=4 must produce off-policy data"
+    # and the correction changes the learned parameters
+    _, p4_gae = run(4, False)
+    assert p4_vt != p4_gae
+    # on-policy (sync=1, lam=1): V-trace reduces to GAE -> identical runs
+    _, p1_vt = run(1, True)
+    _, p1_gae = run(1, False)
+    assert abs(p1_vt - p1_gae) < 1e-5
