@@ -200,10 +200,11 @@ def test_impala_rejects_callback_and_solo():
 def test_gpu_hip_resume_restores_adam_state(tmp_path):
     """End-to-end HIP fast-path resume: a checkpoint
     saved with HipAdam state must feed the lazily-created HipAdam on
-    resume. Asserted via the Adam step counter t in the re-saved
-    checkpoint: 16 hip steps/iter (4 epochs x 4 minibatches), so after
-    2 iters t=32, and after resume+1 more iter t must be 48 (t=16 would
-    mean the state was dropped)."""
+    resume. Asserted via the Adam step counter t: one trained iteration
+    is 16 hip steps (4 epochs x 4 minibatches) and the hipGraph-capture
+    warmup adds 16 more when capture succeeds, so exact counts depend on
+    capture mode — but a RESTORED counter always satisfies
+    t_after >= t_saved + 16, while a dropped one restarts <= 32."""
     cfg = dataclasses.replace(CONFIGS["c2_ppo_mlp"], num_envs=64,
                               rollout_T=8, jobs_per_env=16)
     train(cfg, iterations=2, log_fn=None, out_dir=str(tmp_path),
@@ -212,12 +213,20 @@ def test_gpu_hip_resume_restores_adam_state(tmp_path):
     assert ck.exists()
     meta = load_policy(ck)
     sd = meta["optim_state"]
-    assert sd.get("hip_adam") and int(sd["t"].item()) == 32
+    assert sd.get("hip_adam")
+    t1 = int(sd["t"].item())
+    assert t1 >= 32  # 2 iterations trained (+16 if capture warmup ran)
     train(cfg, iterations=1, log_fn=None, out_dir=str(tmp_path),
           save_every=1, resume=str(ck))
     ck2 = tmp_path / "policy_000003.pt"
     assert ck2.exists()
     sd2 = load_policy(ck2)["optim_state"]
-    assert int(sd2["t"].item()) == 48, \
-        "optimizer state was not restored through the lazy HipAdam path"
+    t2 = int(sd2["t"].item())
+    assert t2 >= t1 + 16, \
+        f"optimizer state dropped through the lazy HipAdam path ({t1}->{t2})"
     assert torch.isfinite(sd2["m"]).all()
+    # bc must track t (bias correction correct post-resume)
+    bc = sd2["bc"]
+    import numpy as np
+    np.testing.assert_allclose(
+        float(bc[0]), 0.9 ** t2, rtol=1e-4)
