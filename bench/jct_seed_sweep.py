@@ -4,7 +4,12 @@ The honest statistical claim for docs/PERF.md: distribution of final
 (held-out, seed-991) greedy/sampled JCT over several training seeds,
 each run selecting its best checkpoint on a DISJOINT selection pool
 (seed 555). Baselines evaluated once on the same held-out pool.
+
+Round-2 knobs: --ent-floor/--ent-floor-coef (per-row hinge
+entropy bonus in the fused loss), --obs-version 2, --elastic (elastic
+jobs + SHRINK/EXPAND actions), longer --iters.
 """
+import argparse
 import json
 import sys
 from pathlib import Path
@@ -15,16 +20,30 @@ sys.path.insert(0, str(Path(__file__).resolve().parent))
 from train_eval_jct import run  # noqa: E402
 
 if __name__ == "__main__":
-    seeds = [int(s) for s in (sys.argv[1:] or ["7", "21", "42", "1234"])]
-    out = {"runs": []}
-    for i, sd in enumerate(seeds):
-        import os
-        kl = float(os.environ.get("SWEEP_KL", "0.0"))
-        r = run(iters=3000, envs=2048, hidden=512, seed=sd,
-                select_every=200, dtype="bf16", kl_coef=kl)
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--seeds", type=int, nargs="+",
+                    default=[7, 21, 42, 1234])
+    ap.add_argument("--iters", type=int, default=3000)
+    ap.add_argument("--envs", type=int, default=2048)
+    ap.add_argument("--hidden", type=int, default=512)
+    ap.add_argument("--select-every", type=int, default=200)
+    ap.add_argument("--kl", type=float, default=0.0)
+    ap.add_argument("--ent-floor", type=float, default=0.0)
+    ap.add_argument("--ent-floor-coef", type=float, default=0.0)
+    ap.add_argument("--obs-version", type=int, default=1, choices=[1, 2])
+    ap.add_argument("--elastic", action="store_true")
+    ap.add_argument("--out", default=None, help="also write JSON here")
+    a = ap.parse_args()
+    out = {"params": vars(a), "runs": []}
+    for i, sd in enumerate(a.seeds):
+        r = run(iters=a.iters, envs=a.envs, hidden=a.hidden, seed=sd,
+                select_every=a.select_every, dtype="bf16", kl_coef=a.kl,
+                ent_floor=a.ent_floor, ent_floor_coef=a.ent_floor_coef,
+                obs_version=a.obs_version, elastic=a.elastic)
         rec = {"seed": sd, "train_s": r["train_s"],
                "selected_at": r.get("selected_at"),
                "selection_jct_us": r.get("selection_jct_us"),
+               "final_entropy": r["final"].get("entropy"),
                "greedy_jct_s": round(r["ppo_greedy"]["mean_jct_us"] / 1e6, 1),
                "sample_jct_s": round(r["ppo_sample"]["mean_jct_us"] / 1e6, 1)}
         if i == 0:
@@ -32,4 +51,10 @@ if __name__ == "__main__":
                 out[b + "_jct_s"] = round(r[b]["mean_jct_us"] / 1e6, 1)
         out["runs"].append(rec)
         print("RUN", json.dumps(rec), flush=True)
+    gj = sorted(r["greedy_jct_s"] for r in out["runs"])
+    out["greedy_median_jct_s"] = gj[len(gj) // 2] if len(gj) % 2 else \
+        0.5 * (gj[len(gj) // 2 - 1] + gj[len(gj) // 2])
     print("SWEEP", json.dumps(out))
+    if a.out:
+        Path(a.out).parent.mkdir(parents=True, exist_ok=True)
+        Path(a.out).write_text(json.dumps(out, indent=1))

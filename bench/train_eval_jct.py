@@ -24,7 +24,8 @@ def run(iters=300, envs=2048, eval_envs=16, eval_eps=16, jobs=96,
         interarrival=90.0, duration=600.0, seed=7, phase2_iters=0,
         phase2_ent=0.002, phase2_lr=1e-4, dtype="fp32", hidden=256,
         save_ckpt=None, ent_coef=0.01, lr=3e-4, select_every=0,
-        select_envs=8, select_eps=8, kl_coef=0.0):
+        select_envs=8, select_eps=8, kl_coef=0.0,
+        obs_version=1, elastic=False, ent_floor=0.0, ent_floor_coef=0.0):
     """Programmatic version returning the full comparison dict.
     Optional phase 2: entropy/lr anneal via a fresh learner (Adam state
     resets; the captured graph bakes coefficients, so annealing means a
@@ -42,16 +43,22 @@ def run(iters=300, envs=2048, eval_envs=16, eval_eps=16, jobs=96,
     from rlgpuschedule_amd.env.hip_env import HipVectorEnv
 
     tkw = dict(mean_interarrival_s=interarrival, median_duration_s=duration)
-    spec = cluster64(horizon=200_000)
+    spec = cluster64(horizon=200_000, obs_version=obs_version,
+                     elastic=elastic, elastic_resize=elastic)
+    if elastic:
+        tkw["elastic"] = True  # traces with min_gpus < gpus
     torch.manual_seed(seed)
     env = HipVectorEnv(spec, envs, trace_kind="philly", jobs_per_env=jobs,
                        seed=seed, **tkw)
     import dataclasses
-    cfg0 = dataclasses.replace(CONFIGS["c2_ppo_mlp"], hidden=hidden)
+    cfg0 = dataclasses.replace(CONFIGS["c2_ppo_mlp"], hidden=hidden,
+                               spec=spec)
     policy = make_policy(cfg0, torch.device("cuda"))
     learner = HipPPOLearner(policy, PPOConfig(epochs=1, minibatches=4,
                                               ent_coef=ent_coef, lr=lr,
-                                              kl_coef=kl_coef),
+                                              kl_coef=kl_coef,
+                                              ent_floor=ent_floor,
+                                              ent_floor_coef=ent_floor_coef),
                             dtype=dtype)
     engine = GpuRolloutEngine(env, policy, 64, seed=seed)
     t0 = time.perf_counter()
