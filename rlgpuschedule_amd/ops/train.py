@@ -211,11 +211,12 @@ class HipAdam:
         if "bc" in sd:
             self.bc.copy_(sd["bc"].to(self.bc.device))
         else:
-            # legacy checkpoint without the running beta^t products:
-            # recompute so bias correction stays right after resume
+            # legacy checkpoint without the bias-correction denominators
+            # (bc = 1 - beta^t, see adam_bump_kernel): recompute from t
+            # so bias correction stays right after resume
             t = float(sd["t"].reshape(-1)[0].item())
             self.bc.copy_(torch.tensor(
-                [self.betas[0] ** t, self.betas[1] ** t],
+                [1.0 - self.betas[0] ** t, 1.0 - self.betas[1] ** t],
                 device=self.bc.device))
 
 

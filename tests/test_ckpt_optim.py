@@ -29,7 +29,7 @@ def test_hipadam_state_roundtrip_includes_bc():
     a.m.fill_(0.25)
     a.v.fill_(0.5)
     a.t.fill_(7)
-    a.bc.copy_(torch.tensor([0.9 ** 7, 0.999 ** 7]))
+    a.bc.copy_(torch.tensor([1 - 0.9 ** 7, 1 - 0.999 ** 7]))
     sd = a.state_dict()
     assert "bc" in sd and sd["hip_adam"]
     b = _tiny_hip_adam()
@@ -44,8 +44,9 @@ def test_hipadam_legacy_state_recomputes_bc():
     legacy = {"hip_adam": True, "m": torch.zeros(6), "v": torch.zeros(6),
               "t": torch.tensor([5.0])}
     a.load_state_dict(legacy)
+    # bc holds the bias-correction DENOMINATORS 1-beta^t (adam_bump_kernel)
     np.testing.assert_allclose(
-        a.bc.numpy(), [0.9 ** 5, 0.999 ** 5], rtol=1e-6)
+        a.bc.numpy(), [1 - 0.9 ** 5, 1 - 0.999 ** 5], rtol=1e-6)
 
 
 def test_hipadam_rejects_torch_state():
@@ -108,7 +109,7 @@ def test_safetensors_optim_roundtrip_hip_adam(tmp_path):
     a.m.uniform_(generator=torch.Generator().manual_seed(1))
     a.v.uniform_()
     a.t.fill_(9)
-    a.bc.copy_(torch.tensor([0.9 ** 9, 0.999 ** 9]))
+    a.bc.copy_(torch.tensor([1 - 0.9 ** 9, 1 - 0.999 ** 9]))
     pol = MLPPolicy(12, 5, 16)
     p = tmp_path / "ck.safetensors"
     save_policy(p, pol, optimizer=a)
@@ -154,7 +155,8 @@ def test_lazy_learner_stash_applies_at_alloc():
     lr = L()
     lr._pending_opt_state = {"hip_adam": True, "m": torch.full((6,), 2.0),
                              "v": torch.zeros(6), "t": torch.tensor([4.0]),
-                             "bc": torch.tensor([0.9 ** 4, 0.999 ** 4])}
+                             "bc": torch.tensor([1 - 0.9 ** 4,
+                                                1 - 0.999 ** 4])}
     lr.opt = _tiny_hip_adam()
     resolve_pending_opt_state(lr)
     assert lr._pending_opt_state is None
@@ -225,8 +227,7 @@ def test_gpu_hip_resume_restores_adam_state(tmp_path):
     assert t2 >= t1 + 16, \
         f"optimizer state dropped through the lazy HipAdam path ({t1}->{t2})"
     assert torch.isfinite(sd2["m"]).all()
-    # bc must track t (bias correction correct post-resume)
+    # bc must track t (bc = 1-beta^t, the bias-correction denominators)
     bc = sd2["bc"]
-    import numpy as np
     np.testing.assert_allclose(
-        float(bc[0]), 0.9 ** t2, rtol=1e-4)
+        float(bc[0]), 1.0 - 0.9 ** t2, rtol=1e-3)
