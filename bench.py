@@ -68,7 +68,8 @@ def _episode_stats(env):
 
 def _gpu_bench(spec, E, T, policy_kind, hidden, actor_dtype, update_dtype,
                minibatches, shards, learner_kind, steps, warmup, seed,
-               trace_kind, trace_kw, jobs_per_env, sync_dist=False):
+               trace_kind, trace_kw, jobs_per_env, sync_dist=False,
+               rollout="twokernel"):
     """Build a fresh env/policy/learner and time `steps` full iterations.
     Returns (env_steps_per_sec_1gpu, ms_per_step, episode_stats)."""
     from rlgpuschedule_amd.agents.ppo_gpu import FusedPPOLearner
@@ -93,7 +94,7 @@ def _gpu_bench(spec, E, T, policy_kind, hidden, actor_dtype, update_dtype,
     else:
         learner = FusedPPOLearner(policy, cfg, grad_sync)
     collector = GpuRolloutEngine(env, policy, T, seed=seed + 1000,
-                                 actor_dtype=actor_dtype)
+                                 actor_dtype=actor_dtype, rollout=rollout)
 
     def one_iter():
         learner.update(collector.collect())
@@ -134,6 +135,10 @@ def main():
     ap.add_argument("--learner", choices=["hip", "fused"], default="hip",
                     help="hip: fully hand-written update kernels; fused: "
                          "torch GEMMs + fused HIP loss")
+    ap.add_argument("--rollout-mode", choices=["twokernel", "mega"],
+                    default="twokernel",
+                    help="mega: fused whole-rollout kernel (fp32 actor, "
+                         "1 shard)")
     ap.add_argument("--trace-file", default=None,
                     help="replay a real Philly-style CSV trace "
                          "(schema: rlgpuschedule_amd/traces/philly.py) "
@@ -166,12 +171,15 @@ def main():
                          seed=97 + rank, **trace_kw), device)
 
     if has_gpu:
+        if args.rollout_mode == "mega":
+            args.actor_dtype = "fp32"
+            args.shards = 1
         value_1gpu, ms_per_step, st = _gpu_bench(
             spec, E, T, args.policy, args.hidden, args.actor_dtype,
             args.update_dtype, args.minibatches, args.shards, args.learner,
             args.steps, args.warmup, seed=97 + rank * 31,
             trace_kind=trace_kind, trace_kw=trace_kw, jobs_per_env=JPE,
-            sync_dist=True)
+            sync_dist=True, rollout=args.rollout_mode)
         n_gpus = world
         value = value_1gpu * n_gpus  # whole-job aggregate (weak scaling)
         elapsed_ms = ms_per_step

@@ -21,7 +21,7 @@ from .rollout import RolloutBuffer
 class GpuRolloutEngine:
     def __init__(self, env, policy, T: int, seed: int = 0,
                  use_graph: bool = True, num_bufs: int = 1,
-                 actor_dtype: str = "fp32"):
+                 actor_dtype: str = "fp32", rollout: str = "twokernel"):
         """`env` may be a LIST of HipVectorEnvs (shards): each shard rolls
         out on its own HIP stream so shard A's env_step overlaps shard B's
         actor kernel (the per-step chain is serial within a shard). The
@@ -33,6 +33,15 @@ class GpuRolloutEngine:
         self.policy = policy
         self.T = T
         self.device = torch.device("cuda")
+        # rollout="mega": the whole T-step rollout is ONE fused kernel
+        # (csrc/rollout_mega.hip) — bit-identical outputs to the
+        # two-kernel graph path (tests/test_mega_rollout.py). MLP + fp32
+        # actor + single shard only.
+        self.rollout_mode = rollout
+        if rollout == "mega":
+            assert not policy.is_recurrent, "mega rollout: MLP policy only"
+            assert actor_dtype == "fp32", "mega rollout: fp32 actor only"
+            assert len(self.envs) == 1, "mega rollout: single shard only"
         # one behavior-policy mirror per buffer: lets an async learner
         # refresh mirror i (after its update) while a rollout on the other
         # parity is still reading mirror i^1 (agents/impala.py)
@@ -131,6 +140,13 @@ class GpuRolloutEngine:
         self.u_all.uniform_(generator=self.gen)
         buf.obs[0].copy_(self.obs_last)
         buf.mask[0].copy_(self.mask_last)
+        if self.rollout_mode == "mega":
+            from ..ops.mega import mega_rollout
+            mega_rollout(self.env, actor, buf, self.u_all,
+                         self.obs_last, self.mask_last,
+                         torch.cuda.current_stream().cuda_stream)
+            self.buf = buf
+            return buf
         nsh = len(self.envs)
         if self.use_graph:
             for shard in range(nsh):

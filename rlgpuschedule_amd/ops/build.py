@@ -22,6 +22,7 @@ _LIBS = {
     "librlsched_nn.so": ["nn_kernels.hip"],
     "librlsched_train.so": ["train_kernels.hip"],
     "librlsched_gru.so": ["gru_train.hip"],
+    "librlsched_mega.so": ["rollout_mega.hip"],
 }
 
 
