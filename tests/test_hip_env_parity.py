@@ -186,3 +186,21 @@ def test_parity_everything_on():
                    elastic_resize=True, obs_version=2)
     _run_parity(spec, jobs_per_env=20, steps=300,
                 mean_interarrival_s=20.0, median_duration_s=150.0)
+
+
+@pytest.mark.gpu
+def test_parity_cluster256_scale():
+    """parity at the config-3 flagship scale
+    (cluster256 = 32 nodes x 8 GPUs, 256 simulated GPUs) — round-1 parity
+    tests stopped at cluster64-scale specs."""
+    from rlgpuschedule_amd.env.spec import cluster256
+    _run_parity(cluster256(horizon=3000), jobs_per_env=48, steps=300,
+                mean_interarrival_s=15.0, median_duration_s=300.0)
+
+
+@pytest.mark.gpu
+def test_parity_cluster1024_poisson():
+    """Config-5 scale: 1024 simulated GPUs, Poisson arrivals."""
+    from rlgpuschedule_amd.env.spec import cluster1024
+    _run_parity(cluster1024(horizon=2000), trace_kind="poisson",
+                jobs_per_env=40, steps=200)
