@@ -27,6 +27,12 @@ if __name__ == "__main__":
     ap.add_argument("--envs", type=int, default=2048)
     ap.add_argument("--hidden", type=int, default=512)
     ap.add_argument("--select-every", type=int, default=200)
+    ap.add_argument("--select-envs", type=int, default=8)
+    ap.add_argument("--select-eps", type=int, default=8)
+    ap.add_argument("--phase2-iters", type=int, default=0,
+                    help="extra iters at --phase2-lr after the main run "
+                         "(fresh learner/capture)")
+    ap.add_argument("--phase2-lr", type=float, default=1e-4)
     ap.add_argument("--kl", type=float, default=0.0)
     ap.add_argument("--ent-floor", type=float, default=0.0)
     ap.add_argument("--ent-floor-coef", type=float, default=0.0)
@@ -39,7 +45,9 @@ if __name__ == "__main__":
         r = run(iters=a.iters, envs=a.envs, hidden=a.hidden, seed=sd,
                 select_every=a.select_every, dtype="bf16", kl_coef=a.kl,
                 ent_floor=a.ent_floor, ent_floor_coef=a.ent_floor_coef,
-                obs_version=a.obs_version, elastic=a.elastic)
+                obs_version=a.obs_version, elastic=a.elastic,
+                select_envs=a.select_envs, select_eps=a.select_eps,
+                phase2_iters=a.phase2_iters, phase2_lr=a.phase2_lr)
         rec = {"seed": sd, "train_s": r["train_s"],
                "selected_at": r.get("selected_at"),
                "selection_jct_us": r.get("selection_jct_us"),

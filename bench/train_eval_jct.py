@@ -85,12 +85,17 @@ def run(iters=300, envs=2048, eval_envs=16, eval_eps=16, jobs=96,
         if select_every and (it + 1) % select_every == 0:
             _select(it + 1)
     if phase2_iters:
+        # low-lr settle phase: fresh learner (Adam state + captured graph
+        # bake the coefficients); selection keeps running
         learner = HipPPOLearner(
             policy, PPOConfig(epochs=1, minibatches=4, lr=phase2_lr,
-                              ent_coef=phase2_ent))
+                              ent_coef=phase2_ent, kl_coef=kl_coef),
+            dtype=dtype)
         for it in range(phase2_iters):
             buf = engine.collect()
             stats = learner.update(buf)
+            if select_every and (it + 1) % select_every == 0:
+                _select(iters + it + 1)
     torch.cuda.synchronize()
     res = {"train_s": round(time.perf_counter() - t0, 1),
            "final": {k: round(v, 4) for k, v in stats.items()}}
