@@ -221,12 +221,19 @@ __global__ __launch_bounds__(256) void mega_rollout_kernel(MegaArgs a) {
   const int Dp = (D + 15) & ~15;
   const int lda_o = Dp + 4, lda_h = H + 4;
 
+  // LDS lifetime aliasing (v2): the actor pipeline only ever needs two
+  // tiles alive at once (obs->h1, h1->h2, h2->head), so h2 reuses the
+  // obs region and head reuses the h1 region. 57 KB -> ~37 KB per
+  // workgroup: occupancy 2 -> 4 workgroups/CU, which doubles the
+  // env-phase wave concurrency (the latency-bound part).
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  float* sobs = (float*)smem;                       // [MROWS][Dp+4] kperm
-  float* h1 = sobs + MROWS * lda_o;                 // [MROWS][H+4] kperm
-  float* h2 = h1 + MROWS * lda_h;                   // [MROWS][H+4] kperm
-  float* shead = h2 + MROWS * lda_h;                // [MROWS][MHEADC+1]
-  int64_t* s_act = (int64_t*)(shead + MROWS * (MHEADC + 1));  // [MROWS]
+  const int r1_sz = MROWS * (lda_o > lda_h ? lda_o : lda_h);
+  float* sobs = (float*)smem;                       // region 1 (kperm)
+  float* h2 = sobs;                                 // region 1 (after L2)
+  float* h1 = sobs + r1_sz;                         // region 2 (kperm)
+  float* shead = h1;                                // region 2 (after head)
+  const int r2_sz = MROWS * (lda_h > (MHEADC + 1) ? lda_h : (MHEADC + 1));
+  int64_t* s_act = (int64_t*)(h1 + r2_sz);          // [MROWS]
   int32_t* s_nf = (int32_t*)(s_act + MROWS);        // [4][RLS_MAX_NODES]
 
   for (int64_t t = 0; t < a.T; ++t) {
@@ -282,8 +289,10 @@ extern "C" int rlsched_mega_rollout(const int64_t* ip, const uint64_t* bufs,
   a.last_value = (float*)x[16];
   const int D = (int)a.p.obs_dim, H = (int)a.H;
   const int Dp = (D + 15) & ~15;
-  size_t lds = sizeof(float) * (MROWS * (Dp + 4) + 2 * MROWS * (H + 4)
-                                + MROWS * (MHEADC + 1))
+  const int lda_o = Dp + 4, lda_h = H + 4;
+  const int r1 = MROWS * (lda_o > lda_h ? lda_o : lda_h);
+  const int r2 = MROWS * (lda_h > (MHEADC + 1) ? lda_h : (MHEADC + 1));
+  size_t lds = sizeof(float) * (r1 + r2)
              + sizeof(int64_t) * MROWS
              + sizeof(int32_t) * 4 * RLS_MAX_NODES;
   if (lds > 64 * 1024) {   // H=512 tile: opt in to the 160 KB LDS budget
