@@ -53,6 +53,8 @@ struct EnvView {
   int64_t job_lo, job_hi, base;
   int32_t *node_free;        // LDS slice during kernel execution
   int32_t *node_free_glb;    // global backing (written back at the end)
+  int32_t *run_valid_glb;    // global backing when run state is in LDS
+  int64_t *run_finish_glb;   //   (run_state_to_lds/run_state_writeback)
   int32_t *pend_job;
   int64_t *pend_override;
   int32_t *run_valid, *run_job, *run_gpus;
@@ -538,6 +540,34 @@ __device__ __forceinline__ void nf_to_lds(EnvView& v, int32_t* s_nf,
 __device__ __forceinline__ void nf_writeback(const EnvView& v, int lane) {
   const int N = (int)v.p->N;
   for (int n = lane; n < N; n += RLS_WAVE) v.node_free_glb[n] = v.node_free[n];
+}
+
+// run_valid/run_finish LDS residency (round 2): the two next_event scans
+// + the completion ballot re-read these R-slot arrays from L2 every
+// step — the hottest dependent loads left after node_free moved to LDS
+// (env_step was 27.8% of device time at the r1 headline). Same redirect
+// pattern as nf_to_lds; all writers go through the EnvView pointers, so
+// semantics are untouched. Wave-private slices, no barriers.
+__device__ __forceinline__ void run_state_to_lds(EnvView& v, int32_t* s_rv,
+                                                 int64_t* s_rf, int lane) {
+  const int R = (int)v.p->R;
+  for (int r = lane; r < R; r += RLS_WAVE) {
+    s_rv[r] = v.run_valid[r];
+    s_rf[r] = v.run_finish[r];
+  }
+  v.run_valid_glb = v.run_valid;
+  v.run_finish_glb = v.run_finish;
+  v.run_valid = s_rv;
+  v.run_finish = s_rf;
+}
+
+__device__ __forceinline__ void run_state_writeback(const EnvView& v,
+                                                    int lane) {
+  const int R = (int)v.p->R;
+  for (int r = lane; r < R; r += RLS_WAVE) {
+    v.run_valid_glb[r] = v.run_valid[r];
+    v.run_finish_glb[r] = v.run_finish[r];
+  }
 }
 
 void unpack(const int64_t* params, const uint64_t* bufs, RlsParams& p, RlsBufs& b) {

@@ -29,15 +29,31 @@
 #include "env_device.h"
 
 namespace {
+// Per-wave dynamic-LDS slices: run_finish[R] (8B), run_valid[R], node_free[N].
+__device__ __forceinline__ char* wave_lds_base(const RlsParams& p, int wid) {
+  extern __shared__ __attribute__((aligned(16))) char dsm[];
+  size_t wave_bytes = ((size_t)p.R * 12 + (size_t)p.N * 4 + 15) & ~(size_t)15;
+  return dsm + (size_t)wid * wave_bytes;
+}
+
+__device__ __forceinline__ size_t env_lds_bytes(const RlsParams& p) {
+  size_t wave_bytes = ((size_t)p.R * 12 + (size_t)p.N * 4 + 15) & ~(size_t)15;
+  return wave_bytes * RLS_ENVS_PER_BLOCK;
+}
+
 __global__ __launch_bounds__(RLS_BLOCK) void env_reset_kernel(RlsParams p, RlsBufs b) {
-  __shared__ int32_t s_nf[RLS_ENVS_PER_BLOCK][RLS_MAX_NODES];
   int lane = threadIdx.x & (RLS_WAVE - 1);
   int wid = threadIdx.x >> 6;
   int64_t e = (int64_t)blockIdx.x * RLS_ENVS_PER_BLOCK + wid;
   if (e >= p.E) return;
+  char* base = wave_lds_base(p, wid);
+  int64_t* s_rf = (int64_t*)base;
+  int32_t* s_rv = (int32_t*)(s_rf + p.R);
+  int32_t* s_nf = s_rv + p.R;
   EnvView v = make_view(p, b, e);
   Sc s;
-  nf_to_lds(v, s_nf[wid], lane);
+  nf_to_lds(v, s_nf, lane);
+  run_state_to_lds(v, s_rv, s_rf, lane);
   reset_env(v, s, lane);
   store_sc(b, e, s);
   b.ep_count[e] = 0;
@@ -49,18 +65,23 @@ __global__ __launch_bounds__(RLS_BLOCK) void env_reset_kernel(RlsParams p, RlsBu
   b.done[e] = 0;
   write_obs_mask(v, s, lane);
   nf_writeback(v, lane);
+  run_state_writeback(v, lane);
 }
 
 __global__ __launch_bounds__(RLS_BLOCK) void env_step_kernel(RlsParams p, RlsBufs b) {
-  __shared__ int32_t s_nf[RLS_ENVS_PER_BLOCK][RLS_MAX_NODES];
   int lane = threadIdx.x & (RLS_WAVE - 1);
   int wid = threadIdx.x >> 6;
   int64_t e = (int64_t)blockIdx.x * RLS_ENVS_PER_BLOCK + wid;
   if (e >= p.E) return;
+  char* base = wave_lds_base(p, wid);
+  int64_t* s_rf = (int64_t*)base;
+  int32_t* s_rv = (int32_t*)(s_rf + p.R);
+  int32_t* s_nf = s_rv + p.R;
   EnvView v = make_view(p, b, e);
   Sc s;
   load_sc(b, e, s);
-  nf_to_lds(v, s_nf[wid], lane);
+  nf_to_lds(v, s_nf, lane);
+  run_state_to_lds(v, s_rv, s_rf, lane);
   int64_t a = b.actions[e];
   bool did = false;
   float reward = 0.0f;
@@ -102,7 +123,9 @@ extern "C" int rlsched_env_reset(const int64_t* params, const uint64_t* bufs,
   RlsParams p; RlsBufs b;
   unpack(params, bufs, p, b);
   dim3 grid((unsigned)((p.E + RLS_ENVS_PER_BLOCK - 1) / RLS_ENVS_PER_BLOCK));
-  hipLaunchKernelGGL(env_reset_kernel, grid, dim3(RLS_BLOCK), 0,
+  size_t wave_bytes = ((size_t)p.R * 12 + (size_t)p.N * 4 + 15) & ~(size_t)15;
+  hipLaunchKernelGGL(env_reset_kernel, grid, dim3(RLS_BLOCK),
+                     wave_bytes * RLS_ENVS_PER_BLOCK,
                      (hipStream_t)stream, p, b);
   return (int)hipGetLastError();
 }
@@ -112,7 +135,9 @@ extern "C" int rlsched_env_step(const int64_t* params, const uint64_t* bufs,
   RlsParams p; RlsBufs b;
   unpack(params, bufs, p, b);
   dim3 grid((unsigned)((p.E + RLS_ENVS_PER_BLOCK - 1) / RLS_ENVS_PER_BLOCK));
-  hipLaunchKernelGGL(env_step_kernel, grid, dim3(RLS_BLOCK), 0,
+  size_t wave_bytes = ((size_t)p.R * 12 + (size_t)p.N * 4 + 15) & ~(size_t)15;
+  hipLaunchKernelGGL(env_step_kernel, grid, dim3(RLS_BLOCK),
+                     wave_bytes * RLS_ENVS_PER_BLOCK,
                      (hipStream_t)stream, p, b);
   return (int)hipGetLastError();
 }
