@@ -114,6 +114,7 @@ __global__ __launch_bounds__(RLS_BLOCK) void env_step_kernel(RlsParams p, RlsBuf
   store_sc(b, e, s);
   write_obs_mask(v, s, lane);
   nf_writeback(v, lane);
+  run_state_writeback(v, lane);
 }
 
 }  // namespace
