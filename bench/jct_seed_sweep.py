@@ -33,6 +33,8 @@ if __name__ == "__main__":
                     help="extra iters at --phase2-lr after the main run "
                          "(fresh learner/capture)")
     ap.add_argument("--phase2-lr", type=float, default=1e-4)
+    ap.add_argument("--eval-envs", type=int, default=16)
+    ap.add_argument("--eval-eps", type=int, default=16)
     ap.add_argument("--kl", type=float, default=0.0)
     ap.add_argument("--ent-floor", type=float, default=0.0)
     ap.add_argument("--ent-floor-coef", type=float, default=0.0)
@@ -47,7 +49,8 @@ if __name__ == "__main__":
                 ent_floor=a.ent_floor, ent_floor_coef=a.ent_floor_coef,
                 obs_version=a.obs_version, elastic=a.elastic,
                 select_envs=a.select_envs, select_eps=a.select_eps,
-                phase2_iters=a.phase2_iters, phase2_lr=a.phase2_lr)
+                phase2_iters=a.phase2_iters, phase2_lr=a.phase2_lr,
+                eval_envs=a.eval_envs, eval_eps=a.eval_eps)
         rec = {"seed": sd, "train_s": r["train_s"],
                "selected_at": r.get("selected_at"),
                "selection_jct_us": r.get("selection_jct_us"),
