@@ -35,6 +35,10 @@ if __name__ == "__main__":
     ap.add_argument("--phase2-lr", type=float, default=1e-4)
     ap.add_argument("--eval-envs", type=int, default=16)
     ap.add_argument("--eval-eps", type=int, default=16)
+    ap.add_argument("--bc-rollouts", type=int, default=0,
+                    help=">0: behavior-clone SJF first (agents/bc.py)")
+    ap.add_argument("--bc-epochs", type=int, default=4)
+    ap.add_argument("--bc-lr", type=float, default=1e-3)
     ap.add_argument("--kl", type=float, default=0.0)
     ap.add_argument("--ent-floor", type=float, default=0.0)
     ap.add_argument("--ent-floor-coef", type=float, default=0.0)
@@ -50,8 +54,11 @@ if __name__ == "__main__":
                 obs_version=a.obs_version, elastic=a.elastic,
                 select_envs=a.select_envs, select_eps=a.select_eps,
                 phase2_iters=a.phase2_iters, phase2_lr=a.phase2_lr,
-                eval_envs=a.eval_envs, eval_eps=a.eval_eps)
+                eval_envs=a.eval_envs, eval_eps=a.eval_eps,
+                bc_rollouts=a.bc_rollouts, bc_epochs=a.bc_epochs,
+                bc_lr=a.bc_lr)
         rec = {"seed": sd, "train_s": r["train_s"],
+               "bc_ce": r.get("bc", {}).get("bc_ce"),
                "selected_at": r.get("selected_at"),
                "selection_jct_us": r.get("selection_jct_us"),
                "final_entropy": r["final"].get("entropy"),

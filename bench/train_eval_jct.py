@@ -25,7 +25,8 @@ def run(iters=300, envs=2048, eval_envs=16, eval_eps=16, jobs=96,
         phase2_ent=0.002, phase2_lr=1e-4, dtype="fp32", hidden=256,
         save_ckpt=None, ent_coef=0.01, lr=3e-4, select_every=0,
         select_envs=8, select_eps=8, kl_coef=0.0,
-        obs_version=1, elastic=False, ent_floor=0.0, ent_floor_coef=0.0):
+        obs_version=1, elastic=False, ent_floor=0.0, ent_floor_coef=0.0,
+        bc_rollouts=0, bc_epochs=4, bc_lr=1e-3):
     """Programmatic version returning the full comparison dict.
     Optional phase 2: entropy/lr anneal via a fresh learner (Adam state
     resets; the captured graph bakes coefficients, so annealing means a
@@ -54,6 +55,17 @@ def run(iters=300, envs=2048, eval_envs=16, eval_eps=16, jobs=96,
     cfg0 = dataclasses.replace(CONFIGS["c2_ppo_mlp"], hidden=hidden,
                                spec=spec)
     policy = make_policy(cfg0, torch.device("cuda"))
+    bc_stats = {}
+    if bc_rollouts:
+        # SJF warm start (agents/bc.py): clone the heuristic + value
+        # warmup on a separate env pool, so PPO fine-tunes FROM the
+        # baseline it must beat
+        from rlgpuschedule_amd.agents.bc import pretrain_bc
+        bc_env = HipVectorEnv(spec, min(envs, 1024), trace_kind="philly",
+                              jobs_per_env=jobs, seed=seed + 50_000, **tkw)
+        bc_stats = pretrain_bc(policy, bc_env, T=64,
+                               n_rollouts=bc_rollouts, epochs=bc_epochs,
+                               lr=bc_lr, device="cuda")
     learner = HipPPOLearner(policy, PPOConfig(epochs=1, minibatches=4,
                                               ent_coef=ent_coef, lr=lr,
                                               kl_coef=kl_coef,
@@ -79,6 +91,8 @@ def run(iters=300, envs=2048, eval_envs=16, eval_eps=16, jobs=96,
                              for k, v in policy.state_dict().items()}
             best["at"] = it
 
+    if select_every and bc_rollouts:
+        _select(0)   # the BC(SJF) policy itself is a selection candidate
     for it in range(iters):
         buf = engine.collect()
         stats = learner.update(buf)
@@ -99,6 +113,8 @@ def run(iters=300, envs=2048, eval_envs=16, eval_eps=16, jobs=96,
     torch.cuda.synchronize()
     res = {"train_s": round(time.perf_counter() - t0, 1),
            "final": {k: round(v, 4) for k, v in stats.items()}}
+    if bc_stats:
+        res["bc"] = bc_stats
     if select_every and best["state"] is not None:
         policy.load_state_dict(best["state"])
         res["selected_at"] = best["at"]
