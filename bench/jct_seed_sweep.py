@@ -39,6 +39,7 @@ if __name__ == "__main__":
                     help=">0: behavior-clone SJF first (agents/bc.py)")
     ap.add_argument("--bc-epochs", type=int, default=4)
     ap.add_argument("--bc-lr", type=float, default=1e-3)
+    ap.add_argument("--bc-ent", type=float, default=0.0)
     ap.add_argument("--kl", type=float, default=0.0)
     ap.add_argument("--ent-floor", type=float, default=0.0)
     ap.add_argument("--ent-floor-coef", type=float, default=0.0)
@@ -56,7 +57,7 @@ if __name__ == "__main__":
                 phase2_iters=a.phase2_iters, phase2_lr=a.phase2_lr,
                 eval_envs=a.eval_envs, eval_eps=a.eval_eps,
                 bc_rollouts=a.bc_rollouts, bc_epochs=a.bc_epochs,
-                bc_lr=a.bc_lr)
+                bc_lr=a.bc_lr, bc_ent=a.bc_ent)
         rec = {"seed": sd, "train_s": r["train_s"],
                "bc_ce": r.get("bc", {}).get("bc_ce"),
                "selected_at": r.get("selected_at"),

@@ -26,7 +26,7 @@ def run(iters=300, envs=2048, eval_envs=16, eval_eps=16, jobs=96,
         save_ckpt=None, ent_coef=0.01, lr=3e-4, select_every=0,
         select_envs=8, select_eps=8, kl_coef=0.0,
         obs_version=1, elastic=False, ent_floor=0.0, ent_floor_coef=0.0,
-        bc_rollouts=0, bc_epochs=4, bc_lr=1e-3):
+        bc_rollouts=0, bc_epochs=4, bc_lr=1e-3, bc_ent=0.0):
     """Programmatic version returning the full comparison dict.
     Optional phase 2: entropy/lr anneal via a fresh learner (Adam state
     resets; the captured graph bakes coefficients, so annealing means a
@@ -65,7 +65,7 @@ def run(iters=300, envs=2048, eval_envs=16, eval_eps=16, jobs=96,
                               jobs_per_env=jobs, seed=seed + 50_000, **tkw)
         bc_stats = pretrain_bc(policy, bc_env, T=64,
                                n_rollouts=bc_rollouts, epochs=bc_epochs,
-                               lr=bc_lr, device="cuda")
+                               lr=bc_lr, ent_coef=bc_ent, device="cuda")
     learner = HipPPOLearner(policy, PPOConfig(epochs=1, minibatches=4,
                                               ent_coef=ent_coef, lr=lr,
                                               kl_coef=kl_coef,

@@ -85,9 +85,15 @@ def collect_sjf_rollouts(env, T: int, n_rollouts: int, device,
 def pretrain_bc(policy, env, T: int = 64, n_rollouts: int = 8,
                 epochs: int = 4, batch: int = 65536, lr: float = 1e-3,
                 gamma: float = 0.99, val_coef: float = 0.5,
+                ent_coef: float = 0.0,
                 device: str | torch.device = "cuda", log_fn=None) -> dict:
     """Clone SJF into `policy` (cross-entropy on masked logits) and warm
-    up the value head (MSE on discounted returns). Returns final losses."""
+    up the value head (MSE on discounted returns). Returns final losses.
+
+    ent_coef>0 clones SOFTLY: the argmax still matches SJF but the
+    distribution keeps entropy, so a subsequent PPO fine-tune starts
+    with live exploration instead of a deterministic policy whose
+    entropy gradient has already vanished."""
     device = torch.device(device)
     obs, mask, acts, ret = collect_sjf_rollouts(env, T, n_rollouts, device,
                                                 gamma)
@@ -102,6 +108,10 @@ def pretrain_bc(policy, env, T: int = 64, n_rollouts: int = 8,
             ce = F.cross_entropy(logits, acts[ix])
             vloss = F.mse_loss(value, ret[ix])
             loss = ce + val_coef * vloss
+            if ent_coef:
+                lp = F.log_softmax(logits, dim=-1)
+                ent = -(lp.exp() * lp.clamp_min(-30.0)).sum(-1).mean()
+                loss = loss - ent_coef * ent
             opt.zero_grad(set_to_none=True)
             loss.backward()
             opt.step()
