@@ -78,8 +78,18 @@ def evaluate_policy(spec: EnvSpec, policy, num_envs: int = 64,
 def evaluate_baseline(spec: EnvSpec, name: str, num_envs: int = 16,
                       min_episodes: int = 16, max_steps: int = 200_000,
                       trace_kind: str = "philly", jobs_per_env: int = 64,
-                      seed: int = 1234, **trace_kw) -> dict:
-    """FIFO/SJF/random baselines on the CPU reference env."""
+                      seed: int = 1234, device: str | None = None,
+                      **trace_kw) -> dict:
+    """FIFO/SJF/random baselines. On a GPU the batched HIP env is used
+    with vectorized torch agent rules (bit-parity env + same
+    deterministic rules => same trajectories as the CPU reference for
+    fifo/sjf), which makes big-pool evals (hundreds of trace windows)
+    cheap; otherwise the CPU reference env."""
+    device = device or ("cuda" if torch.cuda.is_available() else "cpu")
+    if device == "cuda":
+        return _evaluate_baseline_gpu(spec, name, num_envs, min_episodes,
+                                      max_steps, trace_kind, jobs_per_env,
+                                      seed, **trace_kw)
     env = CpuVectorEnv(spec, num_envs, trace_kind=trace_kind,
                        jobs_per_env=jobs_per_env, seed=seed, **trace_kw)
     agent = {"fifo": lambda: FifoAgent(spec.queue_visible),
@@ -97,3 +107,38 @@ def evaluate_baseline(spec: EnvSpec, name: str, num_envs: int = 16,
             "mean_return": env.stats.mean_return(),
             "mean_jct_us": env.stats.mean_jct_us(),
             "completed": env.stats.completed_sum, "steps": steps}
+
+
+@torch.no_grad()
+def _evaluate_baseline_gpu(spec, name, num_envs, min_episodes, max_steps,
+                           trace_kind, jobs_per_env, seed, **trace_kw):
+    from ..env.hip_env import HipVectorEnv
+    from .bc import sjf_actions
+    env = HipVectorEnv(spec, num_envs, trace_kind=trace_kind,
+                       jobs_per_env=jobs_per_env, seed=seed, **trace_kw)
+    env.reset()
+    obs = env.b["obs"]
+    mask = env.action_masks()
+    K = spec.queue_visible
+    gen = torch.Generator(device="cuda")
+    gen.manual_seed(seed)
+    noop = torch.full((num_envs,), K, dtype=torch.int64, device="cuda")
+    steps = 0
+    while steps < max_steps:
+        valid = mask[:, :K].bool()
+        if name == "sjf":
+            acts = sjf_actions(obs, mask, K)
+        elif name == "fifo":
+            first = valid.float().argmax(dim=1)   # first True
+            acts = torch.where(valid.any(dim=1), first, noop)
+        else:   # random: uniform over ALL mask-valid actions (incl. noop+)
+            m = mask.bool().float()
+            acts = torch.multinomial(m, 1, generator=gen).squeeze(1)
+        obs, rew, done, info = env.step(acts)
+        mask = info["action_mask"]
+        steps += 1
+        if steps % 512 == 0 and env.episode_stats()["episodes"] >= min_episodes:
+            break
+    st = env.episode_stats()
+    st["steps"] = steps
+    return st

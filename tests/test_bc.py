@@ -57,3 +57,23 @@ def test_bc_pretrain_learns_sjf():
             obs = torch.as_tensor(o, dtype=torch.float32)
             mask = torch.as_tensor(info["action_mask"]).to(torch.uint8)
     assert agree / tot > 0.85, agree / tot
+
+
+import pytest  # noqa: E402
+
+
+@pytest.mark.gpu
+def test_gpu_baseline_eval_matches_cpu():
+    """The GPU baseline evaluator (HIP env + vectorized rules) must
+    reproduce the CPU reference evaluator exactly for the deterministic
+    baselines (bit-parity env, same rules, same pools)."""
+    from rlgpuschedule_amd.agents.evaluate import evaluate_baseline
+    spec = cluster64()
+    for name in ("sjf", "fifo"):
+        kw = dict(num_envs=8, min_episodes=8, jobs_per_env=24, seed=5,
+                  max_steps=20_000)
+        g = evaluate_baseline(spec, name, device="cuda", **kw)
+        c = evaluate_baseline(spec, name, device="cpu", **kw)
+        assert g["completed"] == c["completed"], name
+        assert g["episodes"] >= 8 and c["episodes"] >= 8
+        assert abs(g["mean_jct_us"] - c["mean_jct_us"]) < 1.0, name
