@@ -46,10 +46,16 @@ if __name__ == "__main__":
     ap.add_argument("--obs-version", type=int, default=1, choices=[1, 2])
     ap.add_argument("--elastic", action="store_true")
     ap.add_argument("--out", default=None, help="also write JSON here")
+    ap.add_argument("--save-dir", default=None,
+                    help="save each seed's selected policy checkpoint here")
     a = ap.parse_args()
     out = {"params": vars(a), "runs": []}
     for i, sd in enumerate(a.seeds):
-        r = run(iters=a.iters, envs=a.envs, hidden=a.hidden, seed=sd,
+        ck = (f"{a.save_dir}/seed{sd}.pt" if a.save_dir else None)
+        if ck:
+            Path(a.save_dir).mkdir(parents=True, exist_ok=True)
+        r = run(save_ckpt=ck,
+                iters=a.iters, envs=a.envs, hidden=a.hidden, seed=sd,
                 select_every=a.select_every, dtype="bf16", kl_coef=a.kl,
                 ent_floor=a.ent_floor, ent_floor_coef=a.ent_floor_coef,
                 obs_version=a.obs_version, elastic=a.elastic,
@@ -64,10 +70,16 @@ if __name__ == "__main__":
                "selection_jct_us": r.get("selection_jct_us"),
                "final_entropy": r["final"].get("entropy"),
                "greedy_jct_s": round(r["ppo_greedy"]["mean_jct_us"] / 1e6, 1),
-               "sample_jct_s": round(r["ppo_sample"]["mean_jct_us"] / 1e6, 1)}
+               "sample_jct_s": round(r["ppo_sample"]["mean_jct_us"] / 1e6, 1),
+               # eval integrity: a policy must not "win" by starving hard
+               # windows out of the completed-episode statistics
+               "greedy_episodes": r["ppo_greedy"]["episodes"],
+               "greedy_completed": r["ppo_greedy"]["completed"]}
         if i == 0:
             for b in ("sjf", "fifo", "random"):
                 out[b + "_jct_s"] = round(r[b]["mean_jct_us"] / 1e6, 1)
+                out[b + "_episodes"] = r[b]["episodes"]
+                out[b + "_completed"] = r[b]["completed"]
         out["runs"].append(rec)
         print("RUN", json.dumps(rec), flush=True)
     gj = sorted(r["greedy_jct_s"] for r in out["runs"])
