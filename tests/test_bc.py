@@ -74,6 +74,10 @@ def test_gpu_baseline_eval_matches_cpu():
                   max_steps=20_000)
         g = evaluate_baseline(spec, name, device="cuda", **kw)
         c = evaluate_baseline(spec, name, device="cpu", **kw)
-        assert g["completed"] == c["completed"], name
+        # stopping rules differ (the GPU evaluator polls episode counts
+        # every 512 steps, so it may replay more episodes of the same
+        # deterministic windows) — PER-EPISODE stats must match exactly
         assert g["episodes"] >= 8 and c["episodes"] >= 8
+        assert g["completed"] / g["episodes"] == 24.0, name
+        assert c["completed"] / c["episodes"] == 24.0, name
         assert abs(g["mean_jct_us"] - c["mean_jct_us"]) < 1.0, name
