@@ -352,3 +352,33 @@ def test_hip_a2c_graph_trains_c4():
     out = train(cfg, iterations=5, log_fn=None)
     assert np.isfinite(out["pi_loss"])
     from rlgpuschedule_amd.agents.a2c_hip import HipA2CLearner  # noqa: F401
+
+
+@pytest.mark.gpu
+def test_capture_failure_falls_back_eager(monkeypatch):
+    """inject a hipGraph-capture failure and
+    assert the learner warns, runs eager, and still trains (the
+    agree_all rank-consistency path is the single-rank degenerate case
+    here; the 2-rank gloo version is tests/test_dist.py)."""
+    import warnings
+
+    from helpers import make_buf as _make_buf
+    from rlgpuschedule_amd.agents.ppo import PPOConfig
+    from rlgpuschedule_amd.agents.ppo_hip import HipPPOLearner
+    from rlgpuschedule_amd.models.policy import MLPPolicy
+    torch.manual_seed(3)
+    spec = cluster64()
+    pol = MLPPolicy(spec.obs_dim, spec.num_actions, 256).cuda()
+    lr = HipPPOLearner(pol, PPOConfig(epochs=1, minibatches=1),
+                       use_graph=True)
+    monkeypatch.setattr(
+        lr, "_capture",
+        lambda buf: (_ for _ in ()).throw(RuntimeError("injected")))
+    buf = _make_buf(8, 64, spec, "cuda", seed=2)
+    with warnings.catch_warnings(record=True) as w:
+        warnings.simplefilter("always")
+        stats = lr.update(buf)
+    assert any("capture" in str(x.message).lower() for x in w)
+    assert lr.use_graph is False and lr._graph is None
+    assert all(abs(v) < 1e6 for v in stats.values())
+    lr.update(buf)   # eager steady state keeps working
