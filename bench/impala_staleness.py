@@ -28,7 +28,7 @@ sys.path.insert(0, str(REPO))
 import torch  # noqa: E402
 
 from rlgpuschedule_amd.agents.impala import ImpalaConfig, train_impala  # noqa: E402
-from rlgpuschedule_amd.env.spec import cluster16, cluster1024  # noqa: E402
+from rlgpuschedule_amd.env.spec import cluster16  # noqa: E402
 from rlgpuschedule_amd.trainer import CONFIGS  # noqa: E402
 
 
