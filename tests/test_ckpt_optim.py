@@ -231,3 +231,37 @@ def test_gpu_hip_resume_restores_adam_state(tmp_path):
     bc = sd2["bc"]
     np.testing.assert_allclose(
         float(bc[0]), 1.0 - 0.9 ** t2, rtol=1e-3)
+
+
+def test_a2c_cpu_resume_roundtrip(tmp_path):
+    """A2C (torch RMSprop) save/resume through the shared routing."""
+    cfg = dataclasses.replace(
+        CONFIGS["c4_gandiva_a2c"], num_envs=2, jobs_per_env=8, rollout_T=8,
+        use_hip_env=False, device="cpu", hidden=16)
+    train(cfg, iterations=2, log_fn=None, out_dir=str(tmp_path),
+          save_every=2)
+    ck = tmp_path / "policy_000002.pt"
+    assert ck.exists()
+    meta = load_policy(ck)
+    assert "optim_state" in meta and "param_groups" in meta["optim_state"]
+    out = train(cfg, iterations=1, log_fn=None, resume=str(ck))
+    assert np.isfinite(out["pi_loss"])
+
+
+@pytest.mark.gpu
+def test_gpu_gru_hip_resume_roundtrip(tmp_path):
+    """GRU HIP learner resume: HipAdam state routed through the lazy
+    path (same machinery as the MLP test, recurrent flagship config)."""
+    cfg = dataclasses.replace(CONFIGS["c3_ppo_gru"], num_envs=64,
+                              rollout_T=8, jobs_per_env=16, hidden=64)
+    train(cfg, iterations=2, log_fn=None, out_dir=str(tmp_path),
+          save_every=2)
+    ck = tmp_path / "policy_000002.pt"
+    assert ck.exists()
+    sd = load_policy(ck)["optim_state"]
+    assert sd.get("hip_adam")
+    t1 = int(sd["t"].item())
+    train(cfg, iterations=1, log_fn=None, out_dir=str(tmp_path),
+          save_every=1, resume=str(ck))
+    sd2 = load_policy(tmp_path / "policy_000003.pt")["optim_state"]
+    assert int(sd2["t"].item()) > t1, "GRU resume dropped optimizer state"
