@@ -434,7 +434,20 @@ __device__ void write_obs_mask(const EnvView& v, const Sc& s, int lane) {
     v.obs[2] = (float)s.run_count / (float)v.p->R;
     v.obs[3] = t_next != RLS_NO_EVENT ? lognorm(t_next - s.clock) : 0.0f;
   }
-  // per-slot features + mask: lane k handles slot k (divergent fits scans)
+  // S-prefix free capacity: first_fit(g) succeeds iff g <= cap_S, the
+  // free total of the FIRST max_span nonzero-free nodes in index order
+  // (first-fit always consumes exactly that prefix — cpu_ref.fits_cap).
+  // One redundant-scalar O(N) pass replaces K per-slot first-fit scans.
+  int capS = 0;
+  {
+    int used = 0;
+    const int S = (int)v.p->S;
+    for (int n = 0; n < N && used < S; ++n) {
+      int f = v.node_free[n];
+      if (f > 0) { capS += f; ++used; }
+    }
+  }
+  // per-slot features + mask: lane k handles slot k
   if (lane < K && !(skip & 1)) {
     bool valid = lane < s.pend_count;
     bool fits = false;
@@ -442,7 +455,7 @@ __device__ void write_obs_mask(const EnvView& v, const Sc& s, int lane) {
       int32_t j = v.pend_job[lane];
       int g = v.trace_gpus[j];
       int g_fit = v.p->elastic ? v.trace_ming[j] : g;
-      fits = first_fit(v, g_fit, nullptr, nullptr) >= 0;
+      fits = g_fit <= capS;
       int bix = 4 + 5 * lane;
       v.obs[bix + 0] = 1.0f;
       v.obs[bix + 1] = (float)g / (float)(v.p->N * v.p->G);

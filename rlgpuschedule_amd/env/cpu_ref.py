@@ -72,6 +72,23 @@ class ClusterEnvRef:
                     return pairs
         return None
 
+    def fits_cap(self) -> int:
+        """Largest g for which first_fit(g) succeeds = the free capacity
+        of the FIRST max_span nonzero-free nodes in index order
+        (first-fit always consumes exactly that node prefix, so
+        first_fit(g) is not None  <=>  g <= fits_cap() for g >= 1).
+        Lets the K per-slot fits bits cost one O(N) pass instead of K."""
+        cap = 0
+        used = 0
+        for n in range(self.spec.num_nodes):
+            f = int(self.node_free[n])
+            if f > 0:
+                cap += f
+                used += 1
+                if used == self.spec.max_span:
+                    break
+        return cap
+
     def _next_event(self) -> int:
         t = NO_EVENT
         for r in range(self.spec.run_cap):
@@ -333,11 +350,12 @@ class ClusterEnvRef:
     def action_mask(self) -> np.ndarray:
         s = self.spec
         m = np.zeros(s.num_actions, bool)
+        cap = self.fits_cap()
         for k in range(min(s.queue_visible, self.pend_count)):
             j = int(self.pend_job[k])
             g_fit = (int(self.trace.min_gpus[j]) if s.elastic
                      else int(self.trace.gpus[j]))
-            m[k] = self.first_fit(g_fit) is not None
+            m[k] = g_fit <= cap
         m[s.noop_action] = True
         if s.gandiva:
             m[s.migrate_action] = self.run_count > 0
@@ -367,6 +385,7 @@ class ClusterEnvRef:
         obs[2] = np.float32(self.run_count) / np.float32(s.run_cap)
         t_next = self._next_event()
         obs[3] = _lognorm(t_next - self.clock) if t_next != NO_EVENT else np.float32(0)
+        cap = self.fits_cap()
         for k in range(min(s.queue_visible, self.pend_count)):
             j = int(self.pend_job[k])
             g = int(self.trace.gpus[j])
@@ -376,8 +395,7 @@ class ClusterEnvRef:
             obs[b + 1] = np.float32(g) / np.float32(C)
             obs[b + 2] = _lognorm(self.clock - self._arrival(j))
             obs[b + 3] = _lognorm(self._pend_work_us(k))
-            obs[b + 4] = np.float32(
-                1.0 if self.first_fit(g_fit) is not None else 0.0)
+            obs[b + 4] = np.float32(1.0 if g_fit <= cap else 0.0)
         if s.obs_version >= 2:
             e = 4 + 5 * s.queue_visible
             gpn = s.gpus_per_node

@@ -394,3 +394,30 @@ def test_elastic_resize_expand_partial_fit():
     assert env.step(spec.expand_action)
     assert int(env.run_gpus[0]) == 5
     env.validate_invariants()
+
+
+def test_fits_cap_equivalent_to_first_fit():
+    """The S-prefix capacity shortcut (fits_cap) must answer fits(g)
+    EXACTLY like the full first-fit scan, for any node_free vector —
+    this equivalence is what lets the obs/mask phase do one O(N) pass
+    instead of K scans (both the oracle and the HIP kernel rely on it)."""
+    import numpy as np
+
+    from rlgpuschedule_amd.env.cpu_ref import ClusterEnvRef
+    from rlgpuschedule_amd.env.spec import EnvSpec
+    from rlgpuschedule_amd.traces.synthetic import philly_like
+    rng = np.random.default_rng(0)
+    for trial in range(300):
+        N = int(rng.integers(1, 20))
+        G = int(rng.integers(1, 9))
+        S = int(rng.integers(1, 9))
+        spec = EnvSpec(num_nodes=N, gpus_per_node=G, max_span=S,
+                       queue_visible=4, queue_cap=8)
+        tr = philly_like(8, seed=trial, max_job_gpus=spec.max_job_gpus)
+        env = ClusterEnvRef(spec, tr, 0, 8)
+        env.reset()
+        env.node_free[:] = rng.integers(0, G + 1, size=N)
+        cap = env.fits_cap()
+        for g in range(1, N * G + 2):
+            assert (env.first_fit(g) is not None) == (g <= cap), \
+                (trial, g, cap, env.node_free)
