@@ -204,3 +204,35 @@ def test_parity_cluster1024_poisson():
     from rlgpuschedule_amd.env.spec import cluster1024
     _run_parity(cluster1024(horizon=2000), trace_kind="poisson",
                 jobs_per_env=40, steps=200)
+
+
+@pytest.mark.gpu
+def test_parity_trace_file_pool():
+    """Real-trace (CSV) pools drive the HIP env bit-identically to the
+    CPU oracle (the loader feeds the same JobTrace SoA both consume)."""
+    from pathlib import Path
+    fix = Path(__file__).parent / "data" / "philly_sample.csv"
+    pool, lo, hi = make_pool("file", 4, 64, seed=11, max_job_gpus=64,
+                             path=str(fix))
+    import torch
+
+    from rlgpuschedule_amd.env.gym_api import CpuVectorEnv
+    from rlgpuschedule_amd.env.hip_env import HipVectorEnv
+    spec = cluster64(horizon=2000)
+    cpu = CpuVectorEnv(spec, 4, pool=pool, job_lo=lo, job_hi=hi)
+    gpu = HipVectorEnv(spec, 4, pool=pool, job_lo=lo, job_hi=hi)
+    obs_c = cpu.reset()
+    obs_g = gpu.reset().cpu().numpy()
+    np.testing.assert_allclose(obs_c, obs_g, atol=1e-5)
+    rng = np.random.default_rng(3)
+    mask_c = cpu.action_masks()
+    for t in range(200):
+        mask_g = gpu.action_masks().cpu().numpy().astype(bool)
+        np.testing.assert_array_equal(mask_c, mask_g, err_msg=f"step {t}")
+        acts = np.array([rng.choice(np.flatnonzero(mask_c[i]))
+                         for i in range(4)], np.int64)
+        obs_c, rew_c, done_c, info_c = cpu.step(acts)
+        obs_g_t, rew_g, done_g, _ = gpu.step(torch.from_numpy(acts).cuda())
+        np.testing.assert_allclose(rew_c, rew_g.cpu().numpy(), atol=1e-5)
+        np.testing.assert_allclose(obs_c, obs_g_t.cpu().numpy(), atol=1e-5)
+        mask_c = info_c["action_mask"]
