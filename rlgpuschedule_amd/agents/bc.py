@@ -41,8 +41,9 @@ def collect_sjf_rollouts(env, T: int, n_rollouts: int, device,
     returns) tensors on `device`. Returns are discounted reward-to-go
     with 0 bootstrap at each rollout end (value-head warmup targets)."""
     K = env.spec.queue_visible
-    is_torch = isinstance(getattr(env, "reset")(), torch.Tensor)
+    is_torch = hasattr(env, "b")   # HipVectorEnv (device buffers)
     if is_torch:
+        env.reset()
         obs = env.b["obs"].clone()
         mask = env.action_masks().clone()
     else:
