@@ -55,6 +55,9 @@ def main(argv=None):
     sw.add_argument("--select-every", type=int, default=200)
     sw.add_argument("--kl-coef", type=float, default=0.2,
                     help="KL-to-behavior penalty (prevents collapse)")
+    sw.add_argument("--bc-rollouts", type=int, default=0,
+                    help=">0: SJF behavior-cloning warm start per seed")
+    sw.add_argument("--bc-ent", type=float, default=0.03)
     sw.add_argument("--save-best", default=None, help="checkpoint path for "
                     "the winning seed's selected policy")
     ls = sub.add_parser("configs", help="list configs")
@@ -107,7 +110,8 @@ def main(argv=None):
         if over:
             cfg = dataclasses.replace(cfg, **over)
         cfg = dataclasses.replace(
-            cfg, ppo=dataclasses.replace(cfg.ppo, kl_coef=args.kl_coef))
+            cfg, ppo=dataclasses.replace(cfg.ppo, kl_coef=args.kl_coef),
+            bc_rollouts=args.bc_rollouts, bc_ent=args.bc_ent)
         res = train_multi_seed(cfg, args.seeds, iterations=args.iterations,
                                select_every=args.select_every,
                                log_fn=print, save_best=args.save_best)

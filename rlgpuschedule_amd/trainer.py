@@ -36,6 +36,12 @@ class TrainConfig:
     trace_kw: dict = field(default_factory=dict)
     actor_dtype: str = "bf16"     # rollout policy compute (fp32 | bf16)
     update_dtype: str = "fp32"    # learner GEMMs (fp32 exact | bf16 mixed)
+    # SJF behavior-cloning warm start (agents/bc.py; round-2 quality
+    # recipe): 0 = off; >0 = clone SJF from this many rollouts before RL
+    bc_rollouts: int = 0
+    bc_epochs: int = 4
+    bc_lr: float = 1e-3
+    bc_ent: float = 0.03          # soft-clone entropy (keeps exploration)
 
 
 CONFIGS = {
@@ -128,6 +134,20 @@ def train(cfg: TrainConfig, iterations: int | None = None,
     policy = make_policy(cfg, device)
     if not solo:
         broadcast_params(policy)
+    if cfg.bc_rollouts and cfg.algo in ("ppo", "a2c"):
+        # SJF warm start on a separate env pool; resume (below) overrides
+        from dataclasses import replace as _replace
+
+        from .agents.bc import pretrain_bc
+        bc_env = make_env(_replace(cfg, num_envs=min(cfg.num_envs, 1024)),
+                          device, rank + 977)
+        if hasattr(bc_env, "venv"):
+            bc_env = bc_env.venv   # bc drives CpuVectorEnv directly
+        pretrain_bc(policy, bc_env, T=cfg.rollout_T,
+                    n_rollouts=cfg.bc_rollouts, epochs=cfg.bc_epochs,
+                    lr=cfg.bc_lr, ent_coef=cfg.bc_ent, device=device)
+        if not solo:
+            broadcast_params(policy)   # ranks cloned from different data
     grad_sync = FlatGradAllreduce(policy) if world > 1 else None
     on_gpu_fast = device.type == "cuda" and cfg.use_hip_env
     if on_gpu_fast:
@@ -173,6 +193,9 @@ def train(cfg: TrainConfig, iterations: int | None = None,
     iters = iterations if iterations is not None else cfg.iterations
     stats = {}
     from .metrics.logger import check_finite
+    if (callback is not None and rank == 0 and cfg.bc_rollouts
+            and start_it == 0):
+        callback(-1, policy, {})   # BC warm start as a selection candidate
     for it in range(start_it, start_it + iters):
         buf = collector.collect()
         if learner is not None:

@@ -81,3 +81,39 @@ def test_gpu_baseline_eval_matches_cpu():
         assert g["completed"] / g["episodes"] == 24.0, name
         assert c["completed"] / c["episodes"] == 24.0, name
         assert abs(g["mean_jct_us"] - c["mean_jct_us"]) < 1.0, name
+
+
+def test_trainer_bc_warm_start_cpu():
+    """TrainConfig.bc_rollouts wires the warm start through
+    trainer.train; the resulting policy is SJF-like before RL took many
+    steps (pi_loss finite, training proceeds)."""
+    import dataclasses
+
+    from rlgpuschedule_amd.trainer import CONFIGS, train
+    cfg = dataclasses.replace(
+        CONFIGS["c2_ppo_mlp"], num_envs=4, jobs_per_env=12, rollout_T=8,
+        use_hip_env=False, device="cpu", hidden=32, bc_rollouts=2,
+        bc_epochs=3)
+    out = train(cfg, iterations=2, log_fn=None)
+    assert np.isfinite(out["pi_loss"])
+
+
+def test_train_sweep_bc_candidate_cpu():
+    """CLI sweep path: with bc_rollouts, the BC warm start is evaluated
+    as the iter-0 selection candidate (selected_at may be 0)."""
+    import dataclasses
+
+    from rlgpuschedule_amd.agents.selection import train_multi_seed
+    from rlgpuschedule_amd.trainer import CONFIGS
+    cfg = dataclasses.replace(
+        CONFIGS["c2_ppo_mlp"], num_envs=4, jobs_per_env=12, rollout_T=8,
+        use_hip_env=False, device="cpu", hidden=32, bc_rollouts=2,
+        bc_epochs=3)
+    cfg = dataclasses.replace(
+        cfg, ppo=dataclasses.replace(cfg.ppo, epochs=1, minibatches=1))
+    res = train_multi_seed(
+        cfg, [1], iterations=2, select_every=2,
+        select_kw=dict(num_envs=2, min_episodes=2, max_steps=2000),
+        eval_kw=dict(num_envs=2, min_episodes=2, max_steps=2000))
+    assert res["runs"][0]["selected_at"] in (0, 2)
+    assert res["best_seed"] == 1
