@@ -117,3 +117,24 @@ def test_train_sweep_bc_candidate_cpu():
         eval_kw=dict(num_envs=2, min_episodes=2, max_steps=2000))
     assert res["runs"][0]["selected_at"] in (0, 2)
     assert res["best_seed"] == 1
+
+
+def test_sjf_actions_random_tensors_match_reference():
+    """sjf_actions vs SjfAgent on RANDOM obs/mask tensors (not just
+    env-reachable states): ties, empty masks, inf-free works."""
+    import numpy as np
+
+    rng = np.random.default_rng(7)
+    K, A = 16, 17
+    ref = SjfAgent(K)
+    for _ in range(200):
+        E = int(rng.integers(1, 9))
+        obs = rng.normal(size=(E, 4 + 5 * K)).astype(np.float32)
+        # quantize work features to force ties
+        obs[:, 7::5] = np.round(obs[:, 7::5] * 2) / 2
+        mask = rng.random((E, A)) < 0.3
+        mask[:, K] = True
+        a_ref = ref.act(obs, mask)
+        a_vec = sjf_actions(torch.from_numpy(obs),
+                            torch.from_numpy(mask.astype(np.uint8)), K)
+        np.testing.assert_array_equal(a_ref, a_vec.numpy())
